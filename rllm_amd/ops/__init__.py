@@ -1,0 +1,274 @@
+"""rllm_amd.ops — CDNA4 HIP kernel library + autograd wrappers.
+
+The `_C` extension (built in-tree by rllm_amd/ops/build.py) is REQUIRED on
+any machine with a GPU: ops fail loudly rather than falling back to eager
+PyTorch, so a silently-degraded bench is impossible. On CPU-only machines
+the module imports fine (for unit tests of the pure-Python layers) but any
+kernel call raises.
+"""
+
+from __future__ import annotations
+
+import importlib.util
+from pathlib import Path
+
+import torch
+
+_C = None
+_LOAD_ERROR: Exception | None = None
+
+
+def _try_load():
+    global _C, _LOAD_ERROR
+    so = Path(__file__).parent / "_C.so"
+    if not so.exists():
+        _LOAD_ERROR = FileNotFoundError(
+            f"{so} not found — build it with `python -m rllm_amd.ops.build` "
+            f"(hipcc --offload-arch=gfx950)"
+        )
+        return
+    try:
+        spec = importlib.util.spec_from_file_location("rllm_amd.ops._C", so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)  # type: ignore[union-attr]
+        _C = mod
+    except Exception as e:  # noqa: BLE001
+        _LOAD_ERROR = e
+
+
+_try_load()
+
+
+def require_ext():
+    """Return the _C extension or raise loudly (never a silent fallback)."""
+    if _C is None:
+        raise RuntimeError(
+            f"rllm_amd HIP extension is not available: {_LOAD_ERROR}. "
+            f"Refusing to fall back to eager PyTorch on the GPU path."
+        ) from _LOAD_ERROR
+    return _C
+
+
+def ext_available() -> bool:
+    return _C is not None
+
+
+# ---------------------------------------------------------------------------
+# Autograd wrappers
+# ---------------------------------------------------------------------------
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, eps: float):
+        C = require_ext()
+        y, inv_rms = C.rmsnorm_fwd(x, weight, eps, True)
+        ctx.save_for_backward(x, weight, inv_rms)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = require_ext()
+        x, weight, inv_rms = ctx.saved_tensors
+        dx, dw = C.rmsnorm_bwd(dy.contiguous(), x, weight, inv_rms)
+        return dx, dw.to(weight.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.requires_grad or weight.requires_grad:
+        return _RMSNormFn.apply(x.contiguous(), weight, eps)
+    C = require_ext()
+    return C.rmsnorm_fwd(x.contiguous(), weight, eps, False)[0]
+
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gateup: torch.Tensor):
+        C = require_ext()
+        out = C.swiglu_fwd(gateup)
+        ctx.save_for_backward(gateup)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        C = require_ext()
+        (gateup,) = ctx.saved_tensors
+        return C.swiglu_bwd(dout.contiguous(), gateup)
+
+
+def swiglu(gateup: torch.Tensor) -> torch.Tensor:
+    """gateup [..., 2I] (gate | up) -> silu(gate) * up, [..., I]."""
+    if gateup.requires_grad:
+        return _SwiGLUFn.apply(gateup.contiguous())
+    return require_ext().swiglu_fwd(gateup.contiguous())
+
+
+class _RoPEFn(torch.autograd.Function):
+    """Out-of-place autograd wrapper over the in-place rope kernel."""
+
+    @staticmethod
+    def forward(ctx, q, k, cos_tab, sin_tab, positions):
+        C = require_ext()
+        q = q.contiguous().clone()
+        k = k.contiguous().clone()
+        C.rope_inplace(q, k, cos_tab, sin_tab, positions, False)
+        ctx.save_for_backward(cos_tab, sin_tab, positions)
+        return q, k
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        C = require_ext()
+        cos_tab, sin_tab, positions = ctx.saved_tensors
+        dq = dq.contiguous().clone()
+        dk = dk.contiguous().clone()
+        C.rope_inplace(dq, dk, cos_tab, sin_tab, positions, True)
+        return dq, dk, None, None, None
+
+
+def rope(q, k, cos_tab, sin_tab, positions):
+    """Apply rotary embedding. q [T,Hq,D], k [T,Hk,D], positions [T] int32."""
+    if q.requires_grad or k.requires_grad:
+        return _RoPEFn.apply(q, k, cos_tab, sin_tab, positions)
+    C = require_ext()
+    q = q.contiguous()
+    k = k.contiguous()
+    C.rope_inplace(q, k, cos_tab, sin_tab, positions, False)
+    return q, k
+
+
+def build_rope_tables(max_pos: int, head_dim: int, theta: float, device) -> tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed cos/sin tables [max_pos, head_dim/2] fp32 (guide: keep
+    trig off the device for memory-bound RoPE)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64) / half))
+    pos = torch.arange(max_pos, dtype=torch.float64)
+    freqs = torch.outer(pos, inv_freq)
+    return freqs.cos().float().to(device), freqs.sin().float().to(device)
+
+
+class _ChunkedLogprobFn(torch.autograd.Function):
+    """Fused token-logprob (+ entropy metric) over hidden @ lm_head^T without
+    materializing [T, vocab] (K7/K12). GEMM chunks via hipBLASLt; online-LSE
+    and dlogits formation are HIP kernels."""
+
+    @staticmethod
+    def forward(ctx, hidden: torch.Tensor, lm_weight: torch.Tensor, targets: torch.Tensor,
+                chunk: int = 16384, temperature: float = 1.0, want_entropy: bool = True):
+        C = require_ext()
+        T = hidden.shape[0]
+        V = lm_weight.shape[0]
+        inv_temp = 1.0 / temperature
+        m = torch.full((T,), -float("inf"), device=hidden.device, dtype=torch.float32)
+        s = torch.zeros(T, device=hidden.device, dtype=torch.float32)
+        e = torch.zeros(T, device=hidden.device, dtype=torch.float32) if want_entropy else None
+        tgt_logit = torch.zeros(T, device=hidden.device, dtype=torch.float32)
+        targets_i32 = targets.to(torch.int32)
+        for c0 in range(0, V, chunk):
+            c1 = min(V, c0 + chunk)
+            logits = torch.matmul(hidden, lm_weight[c0:c1].t()).contiguous()
+            C.lse_chunk_update(logits, m, s, e, tgt_logit, targets_i32, c0, inv_temp)
+        lse = m + torch.log(s)
+        logprob = tgt_logit - lse
+        entropy = (lse - e / s) if want_entropy else None
+        ctx.save_for_backward(hidden, lm_weight, targets_i32, lse)
+        ctx.chunk = chunk
+        ctx.inv_temp = inv_temp
+        return logprob, entropy
+
+    @staticmethod
+    def backward(ctx, dlogprob, dentropy):
+        C = require_ext()
+        hidden, lm_weight, targets_i32, lse = ctx.saved_tensors
+        V = lm_weight.shape[0]
+        chunk = ctx.chunk
+        dlp = dlogprob.contiguous().to(torch.float32)
+        dhidden = torch.zeros_like(hidden)
+        dweight = torch.zeros_like(lm_weight) if lm_weight.requires_grad else None
+        for c0 in range(0, V, chunk):
+            c1 = min(V, c0 + chunk)
+            w_chunk = lm_weight[c0:c1]
+            logits = torch.matmul(hidden, w_chunk.t()).contiguous()
+            dlogits = C.ce_bwd_chunk(logits, lse, dlp, targets_i32, c0, ctx.inv_temp)
+            dhidden += torch.matmul(dlogits, w_chunk)
+            if dweight is not None:
+                dweight[c0:c1] = torch.matmul(dlogits.t(), hidden)
+        return dhidden, dweight, None, None, None, None
+
+
+def chunked_logprob(hidden, lm_weight, targets, chunk: int = 16384,
+                    temperature: float = 1.0, want_entropy: bool = True):
+    """logprob[t] = log softmax(hidden[t] @ W^T / temp)[targets[t]].
+
+    Returns (logprob [T] fp32, entropy [T] fp32 or None). Entropy is a
+    metric only — no gradient flows through it.
+    """
+    return _ChunkedLogprobFn.apply(hidden, lm_weight, targets, chunk, temperature, want_entropy)
+
+
+class _GRPOLossFn(torch.autograd.Function):
+    """Per-token ratio-clip + KL loss; gradient computed in the same fused
+    forward kernel (K8-K10)."""
+
+    @staticmethod
+    def forward(ctx, logprob, old_logprob, ref_logprob, advantages, tis_w,
+                eps_lo: float, eps_hi: float, kl_beta: float):
+        C = require_ext()
+        loss_tok, dlp_tok, clipped = C.grpo_loss_fwd(
+            logprob.contiguous(), old_logprob.contiguous(),
+            ref_logprob.contiguous() if ref_logprob is not None else None,
+            advantages.contiguous(),
+            tis_w.contiguous() if tis_w is not None else None,
+            eps_lo, eps_hi, kl_beta)
+        ctx.save_for_backward(dlp_tok)
+        ctx.mark_non_differentiable(clipped)
+        return loss_tok, clipped
+
+    @staticmethod
+    def backward(ctx, dloss, dclipped):
+        (dlp_tok,) = ctx.saved_tensors
+        return dloss * dlp_tok, None, None, None, None, None, None, None
+
+
+def grpo_loss_per_token(logprob, old_logprob, ref_logprob, advantages, tis_w=None,
+                        eps_lo: float = 0.2, eps_hi: float | None = None, kl_beta: float = 0.0):
+    """Returns (loss_tok [N], clipped [N]) — aggregate with masked mean per
+    loss_agg_mode upstream."""
+    if eps_hi is None:
+        eps_hi = eps_lo
+    return _GRPOLossFn.apply(logprob, old_logprob, ref_logprob, advantages, tis_w,
+                             eps_lo, eps_hi, kl_beta)
+
+
+# ---------------------------------------------------------------------------
+# Raw kernel pass-throughs (inference path, no autograd)
+# ---------------------------------------------------------------------------
+
+
+def flash_prefill(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale):
+    return require_ext().flash_prefill(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale)
+
+
+def paged_decode(q, k_pages, v_pages, block_tables, seq_lens, scale, n_splits: int = 0):
+    return require_ext().paged_decode(q, k_pages, v_pages, block_tables, seq_lens, scale, n_splits)
+
+
+def reshape_and_cache(k, v, k_pages, v_pages, slot_mapping):
+    return require_ext().reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
+
+
+def sample_logprob(logits, temperature: float, seed: int, step: int):
+    return require_ext().sample_logprob(logits, temperature, seed, step)
+
+
+def gather_logprob(logits, tokens, temperature: float = 1.0):
+    return require_ext().gather_logprob(logits, tokens, temperature)
+
+
+def adamw_step(grad, master, m, v, param, gnorm_sq, *, lr, beta1=0.9, beta2=0.999,
+               eps=1e-8, weight_decay=0.0, step, grad_clip=0.0, grad_scale=1.0):
+    require_ext().adamw_step(grad, master, m, v, param, gnorm_sq, lr, beta1, beta2,
+                             eps, weight_decay, step, grad_clip, grad_scale)
+
+
+def grad_sq_sum(grad, grad_scale: float = 1.0):
+    return require_ext().grad_sq_sum(grad, grad_scale)

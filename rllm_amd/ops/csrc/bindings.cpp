@@ -1,0 +1,63 @@
+// Python bindings for the rllm_amd CDNA4 kernel library (_C extension).
+#include <torch/extension.h>
+
+// elementwise.hip
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps, bool save_inv_rms);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w, torch::Tensor inv_rms);
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_tab, torch::Tensor sin_tab,
+                  torch::Tensor positions, bool backward);
+torch::Tensor swiglu_fwd(torch::Tensor gateup);
+torch::Tensor swiglu_bwd(torch::Tensor dout, torch::Tensor gateup);
+
+// logprob.hip
+void lse_chunk_update(torch::Tensor logits, torch::Tensor m_state, torch::Tensor s_state,
+                      c10::optional<torch::Tensor> e_state, torch::Tensor target_logit,
+                      torch::Tensor targets, int64_t chunk_start, double inv_temp);
+torch::Tensor ce_bwd_chunk(torch::Tensor logits, torch::Tensor lse, torch::Tensor dlp,
+                           torch::Tensor targets, int64_t chunk_start, double inv_temp);
+
+// grpo.hip
+std::vector<torch::Tensor> grpo_loss_fwd(torch::Tensor logprob, torch::Tensor old_logprob,
+                                         c10::optional<torch::Tensor> ref_logprob, torch::Tensor advantages,
+                                         c10::optional<torch::Tensor> tis_w,
+                                         double eps_lo, double eps_hi, double kl_beta);
+
+// adamw.hip
+torch::Tensor grad_sq_sum(torch::Tensor grad, double grad_scale);
+void adamw_step(torch::Tensor grad, torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                torch::Tensor param, c10::optional<torch::Tensor> gnorm_sq,
+                double lr, double beta1, double beta2, double eps, double weight_decay,
+                int64_t step, double grad_clip, double grad_scale);
+
+// attention.hip
+torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                            torch::Tensor tile_seq_start, torch::Tensor tile_row0,
+                            torch::Tensor tile_seq_len, double scale);
+torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor v_pages,
+                           torch::Tensor block_tables, torch::Tensor seq_lens, double scale,
+                           int64_t n_splits);
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,
+                       torch::Tensor v_pages, torch::Tensor slot_mapping);
+
+// sampling.hip
+std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
+                                          int64_t seed, int64_t step);
+torch::Tensor gather_logprob(torch::Tensor logits, torch::Tensor tokens, double temperature);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16)");
+  m.def("rope_inplace", &rope_inplace, "Rotary embedding in-place (bf16)");
+  m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (bf16)");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (bf16)");
+  m.def("lse_chunk_update", &lse_chunk_update, "Online LSE + target gather over a vocab chunk");
+  m.def("ce_bwd_chunk", &ce_bwd_chunk, "Chunked logprob backward: dlogits");
+  m.def("grpo_loss_fwd", &grpo_loss_fwd, "Fused GRPO ratio-clip + k3 KL loss (loss, dlp, clipfrac)");
+  m.def("grad_sq_sum", &grad_sq_sum, "Sum of squared gradients (bf16 flat)");
+  m.def("adamw_step", &adamw_step, "Fused AdamW over flat buffers");
+  m.def("flash_prefill", &flash_prefill, "Flash causal GQA prefill (varlen, MFMA)");
+  m.def("paged_decode", &paged_decode, "Paged decode attention (flash-decoding splits)");
+  m.def("reshape_and_cache", &reshape_and_cache, "Scatter K/V into KV pages");
+  m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
+  m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
+}

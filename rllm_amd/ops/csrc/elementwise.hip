@@ -1,0 +1,304 @@
+// Fused elementwise / normalization kernels for the Qwen-family decoder:
+// RMSNorm (fwd+bwd), rotary embedding (fwd; bwd = inverse rotation),
+// SwiGLU (fwd+bwd). All bf16 I/O with fp32 math, vectorized 8-wide.
+//
+// Replaces what the reference delegates to vLLM/flash-attn fused layer ops
+// (SURVEY.md §2.E K5).
+
+#include "common.hpp"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+// ---------------------------------------------------------------------------
+// RMSNorm
+// ---------------------------------------------------------------------------
+
+// One block per row. H must be a multiple of 8 (true for all Qwen configs).
+__global__ void rmsnorm_fwd_kernel(
+    const uint16_t* __restrict__ x,  // [T, H] bf16
+    const uint16_t* __restrict__ w,  // [H] bf16
+    uint16_t* __restrict__ y,        // [T, H] bf16
+    float* __restrict__ inv_rms_out, // [T] f32 (saved for backward; may be null)
+    int H, float eps) {
+  __shared__ float scratch[16];
+  const int64_t row = blockIdx.x;
+  const uint16_t* xr = x + row * (int64_t)H;
+  uint16_t* yr = y + row * (int64_t)H;
+
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    short4v a = *reinterpret_cast<const short4v*>(xr + i);
+    short4v b = *reinterpret_cast<const short4v*>(xr + i + 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float va = bf16_to_f32((uint16_t)a[j]);
+      float vb = bf16_to_f32((uint16_t)b[j]);
+      ss += va * va + vb * vb;
+    }
+  }
+  ss = block_reduce_sum(ss, scratch);
+  const float inv_rms = rsqrtf(ss / (float)H + eps);
+  if (threadIdx.x == 0 && inv_rms_out != nullptr) inv_rms_out[row] = inv_rms;
+
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    short4v a = *reinterpret_cast<const short4v*>(xr + i);
+    short4v b = *reinterpret_cast<const short4v*>(xr + i + 4);
+    short4v wa = *reinterpret_cast<const short4v*>(w + i);
+    short4v wb = *reinterpret_cast<const short4v*>(w + i + 4);
+    short4v oa, ob;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      oa[j] = (short)f32_to_bf16(bf16_to_f32((uint16_t)a[j]) * inv_rms * bf16_to_f32((uint16_t)wa[j]));
+      ob[j] = (short)f32_to_bf16(bf16_to_f32((uint16_t)b[j]) * inv_rms * bf16_to_f32((uint16_t)wb[j]));
+    }
+    *reinterpret_cast<short4v*>(yr + i) = oa;
+    *reinterpret_cast<short4v*>(yr + i + 4) = ob;
+  }
+}
+
+// dx = inv_rms * (dy*w - x * inv_rms^2 * mean_h(dy*w*x))
+__global__ void rmsnorm_bwd_dx_kernel(
+    const uint16_t* __restrict__ dy, // [T, H]
+    const uint16_t* __restrict__ x,  // [T, H]
+    const uint16_t* __restrict__ w,  // [H]
+    const float* __restrict__ inv_rms, // [T]
+    uint16_t* __restrict__ dx,       // [T, H]
+    int H) {
+  __shared__ float scratch[16];
+  const int64_t row = blockIdx.x;
+  const uint16_t* dyr = dy + row * (int64_t)H;
+  const uint16_t* xr = x + row * (int64_t)H;
+  uint16_t* dxr = dx + row * (int64_t)H;
+  const float ir = inv_rms[row];
+
+  float dot = 0.f;
+  for (int i = threadIdx.x * 4; i < H; i += blockDim.x * 4) {
+    short4v d = *reinterpret_cast<const short4v*>(dyr + i);
+    short4v a = *reinterpret_cast<const short4v*>(xr + i);
+    short4v ww = *reinterpret_cast<const short4v*>(w + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      dot += bf16_to_f32((uint16_t)d[j]) * bf16_to_f32((uint16_t)ww[j]) * bf16_to_f32((uint16_t)a[j]);
+  }
+  dot = block_reduce_sum(dot, scratch);
+  const float c = dot * ir * ir / (float)H;
+
+  for (int i = threadIdx.x * 4; i < H; i += blockDim.x * 4) {
+    short4v d = *reinterpret_cast<const short4v*>(dyr + i);
+    short4v a = *reinterpret_cast<const short4v*>(xr + i);
+    short4v ww = *reinterpret_cast<const short4v*>(w + i);
+    short4v o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float g = bf16_to_f32((uint16_t)d[j]) * bf16_to_f32((uint16_t)ww[j]);
+      float xv = bf16_to_f32((uint16_t)a[j]);
+      o[j] = (short)f32_to_bf16(ir * (g - xv * c));
+    }
+    *reinterpret_cast<short4v*>(dxr + i) = o;
+  }
+}
+
+// dw[h] = sum_t dy[t,h] * x[t,h] * inv_rms[t]  (fp32 accumulation)
+__global__ void rmsnorm_bwd_dw_kernel(
+    const uint16_t* __restrict__ dy,
+    const uint16_t* __restrict__ x,
+    const float* __restrict__ inv_rms,
+    float* __restrict__ dw,          // [H] fp32, pre-zeroed
+    int64_t T, int H) {
+  const int h = blockIdx.x * blockDim.x + threadIdx.x;
+  if (h >= H) return;
+  float acc = 0.f;
+  for (int64_t t = blockIdx.y; t < T; t += gridDim.y) {
+    acc += bf16_to_f32(dy[t * H + h]) * bf16_to_f32(x[t * H + h]) * inv_rms[t];
+  }
+  atomicAdd(&dw[h], acc);
+}
+
+// ---------------------------------------------------------------------------
+// Rotary embedding (NeoX rotate-half, as used by the Qwen2 family)
+// ---------------------------------------------------------------------------
+
+// In-place on q [T, Hq, D] and k [T, Hk, D]; cos/sin tables [max_pos, D/2] f32
+// precomputed on host (on-device trig turns this memory-bound op VALU-bound).
+// backward: apply inverse rotation (negated sin).
+__global__ void rope_kernel(
+    uint16_t* __restrict__ q,
+    uint16_t* __restrict__ k,
+    const float* __restrict__ cos_tab,
+    const float* __restrict__ sin_tab,
+    const int32_t* __restrict__ positions, // [T]
+    int64_t T, int Hq, int Hk, int D, int neg_sin) {
+  const int half = D / 2;
+  const int64_t total = T * (int64_t)(Hq + Hk) * half;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int i = (int)(idx % half);
+    int64_t th = idx / half;
+    int h = (int)(th % (Hq + Hk));
+    int64_t t = th / (Hq + Hk);
+
+    uint16_t* base;
+    if (h < Hq) base = q + (t * Hq + h) * D;
+    else        base = k + (t * Hk + (h - Hq)) * D;
+
+    const int pos = positions[t];
+    float c = cos_tab[(int64_t)pos * half + i];
+    float s = sin_tab[(int64_t)pos * half + i];
+    if (neg_sin) s = -s;
+
+    float a = bf16_to_f32(base[i]);
+    float b = bf16_to_f32(base[i + half]);
+    base[i]        = f32_to_bf16(a * c - b * s);
+    base[i + half] = f32_to_bf16(b * c + a * s);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU
+// ---------------------------------------------------------------------------
+
+// gateup [T, 2I] (gate | up concatenated along the feature dim) -> out [T, I]
+__global__ void swiglu_fwd_kernel(
+    const uint16_t* __restrict__ gateup,
+    uint16_t* __restrict__ out,
+    int64_t T, int I) {
+  const int64_t total = T * (int64_t)I / 4;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = idx / (I / 4);
+    int i = (int)(idx % (I / 4)) * 4;
+    const uint16_t* g = gateup + t * 2 * I + i;
+    const uint16_t* u = gateup + t * 2 * I + I + i;
+    short4v gv = *reinterpret_cast<const short4v*>(g);
+    short4v uv = *reinterpret_cast<const short4v*>(u);
+    short4v o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gf = bf16_to_f32((uint16_t)gv[j]);
+      float uf = bf16_to_f32((uint16_t)uv[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      o[j] = (short)f32_to_bf16(gf * sig * uf);
+    }
+    *reinterpret_cast<short4v*>(out + t * I + i) = o;
+  }
+}
+
+__global__ void swiglu_bwd_kernel(
+    const uint16_t* __restrict__ dout,   // [T, I]
+    const uint16_t* __restrict__ gateup, // [T, 2I]
+    uint16_t* __restrict__ dgateup,      // [T, 2I]
+    int64_t T, int I) {
+  const int64_t total = T * (int64_t)I / 4;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = idx / (I / 4);
+    int i = (int)(idx % (I / 4)) * 4;
+    short4v gv = *reinterpret_cast<const short4v*>(gateup + t * 2 * I + i);
+    short4v uv = *reinterpret_cast<const short4v*>(gateup + t * 2 * I + I + i);
+    short4v dv = *reinterpret_cast<const short4v*>(dout + t * I + i);
+    short4v dg, du;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gf = bf16_to_f32((uint16_t)gv[j]);
+      float uf = bf16_to_f32((uint16_t)uv[j]);
+      float df = bf16_to_f32((uint16_t)dv[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      float dsilu = sig * (1.f + gf * (1.f - sig));
+      dg[j] = (short)f32_to_bf16(df * uf * dsilu);
+      du[j] = (short)f32_to_bf16(df * silu);
+    }
+    *reinterpret_cast<short4v*>(dgateup + t * 2 * I + i) = dg;
+    *reinterpret_cast<short4v*>(dgateup + t * 2 * I + I + i) = du;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps, bool save_inv_rms) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dtype() == torch::kBFloat16 && w.is_contiguous());
+  const int64_t T = x.numel() / x.size(-1);
+  const int H = (int)x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  auto y = torch::empty_like(x);
+  torch::Tensor inv_rms;
+  float* inv_ptr = nullptr;
+  if (save_inv_rms) {
+    inv_rms = torch::empty({T}, x.options().dtype(torch::kFloat32));
+    inv_ptr = inv_rms.data_ptr<float>();
+  }
+  dim3 grid((unsigned)T);
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const uint16_t*)x.data_ptr(), (const uint16_t*)w.data_ptr(),
+                     (uint16_t*)y.data_ptr(), inv_ptr, H, (float)eps);
+  HIP_CHECK_KERNEL();
+  if (save_inv_rms) return {y, inv_rms};
+  return {y};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w, torch::Tensor inv_rms) {
+  const int64_t T = x.numel() / x.size(-1);
+  const int H = (int)x.size(-1);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(rmsnorm_bwd_dx_kernel, dim3((unsigned)T), dim3(256), 0, cur_stream(),
+                     (const uint16_t*)dy.data_ptr(), (const uint16_t*)x.data_ptr(),
+                     (const uint16_t*)w.data_ptr(), inv_rms.data_ptr<float>(),
+                     (uint16_t*)dx.data_ptr(), H);
+  HIP_CHECK_KERNEL();
+  dim3 grid2((H + 255) / 256, 128);
+  hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel, grid2, dim3(256), 0, cur_stream(),
+                     (const uint16_t*)dy.data_ptr(), (const uint16_t*)x.data_ptr(),
+                     inv_rms.data_ptr<float>(), dw.data_ptr<float>(), T, H);
+  HIP_CHECK_KERNEL();
+  return {dx, dw};
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_tab, torch::Tensor sin_tab,
+                  torch::Tensor positions, bool backward) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(k.is_cuda() && k.dtype() == torch::kBFloat16 && k.is_contiguous());
+  TORCH_CHECK(positions.dtype() == torch::kInt32);
+  const int64_t T = q.size(0);
+  const int Hq = (int)q.size(1), Hk = (int)k.size(1), D = (int)q.size(2);
+  TORCH_CHECK(k.size(0) == T && k.size(2) == D);
+  const int64_t total = T * (int64_t)(Hq + Hk) * (D / 2);
+  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total, 256)), dim3(256), 0, cur_stream(),
+                     (uint16_t*)q.data_ptr(), (uint16_t*)k.data_ptr(),
+                     cos_tab.data_ptr<float>(), sin_tab.data_ptr<float>(),
+                     positions.data_ptr<int32_t>(), T, Hq, Hk, D, backward ? 1 : 0);
+  HIP_CHECK_KERNEL();
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor gateup) {
+  TORCH_CHECK(gateup.is_cuda() && gateup.dtype() == torch::kBFloat16 && gateup.is_contiguous());
+  const int64_t T = gateup.numel() / gateup.size(-1);
+  const int I2 = (int)gateup.size(-1);
+  TORCH_CHECK(I2 % 8 == 0);
+  const int I = I2 / 2;
+  auto sizes = gateup.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gateup.options());
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for(T * I / 4, 256)), dim3(256), 0, cur_stream(),
+                     (const uint16_t*)gateup.data_ptr(), (uint16_t*)out.data_ptr(), T, I);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor dout, torch::Tensor gateup) {
+  const int64_t T = gateup.numel() / gateup.size(-1);
+  const int I = (int)gateup.size(-1) / 2;
+  auto dgateup = torch::empty_like(gateup);
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for(T * I / 4, 256)), dim3(256), 0, cur_stream(),
+                     (const uint16_t*)dout.data_ptr(), (const uint16_t*)gateup.data_ptr(),
+                     (uint16_t*)dgateup.data_ptr(), T, I);
+  HIP_CHECK_KERNEL();
+  return dgateup;
+}

@@ -1,0 +1,314 @@
+"""GPU numerics tests: every HIP kernel vs its plain-PyTorch fp32 reference.
+
+Tolerances account for bf16 I/O (abs ~1e-2 relative to value scale).
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from rllm_amd import ops
+    from rllm_amd.ops import reference as ref
+else:
+    ops = None
+    ref = None
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+DEV = "cuda"
+
+
+def rand_bf16(*shape, scale=1.0):
+    return (torch.randn(*shape, device=DEV, dtype=torch.float32) * scale).to(torch.bfloat16)
+
+
+@requires_gpu
+def test_rmsnorm_fwd_bwd():
+    torch.manual_seed(0)
+    x = rand_bf16(257, 1536)
+    w = rand_bf16(1536, scale=0.5) + 1.0
+    y = ops.rmsnorm(x, w)
+    y_ref = ref.rmsnorm_ref(x, w)
+    assert (y.float() - y_ref.float()).abs().max() < 2e-2
+
+    # backward
+    x2 = x.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    out = ops.rmsnorm(x2, w2)
+    g = torch.randn_like(out, dtype=torch.float32).to(torch.bfloat16)
+    out.backward(g)
+
+    x3 = x.float().clone().requires_grad_(True)
+    w3 = w.float().clone().requires_grad_(True)
+    inv = torch.rsqrt(x3.pow(2).mean(-1, keepdim=True) + 1e-6)
+    (x3 * inv * w3).backward(g.float())
+    assert (x2.grad.float() - x3.grad).abs().max() < 5e-2
+    assert (w2.grad.float() - w3.grad).abs().max() / w3.grad.abs().max() < 3e-2
+
+
+@requires_gpu
+def test_rope_fwd_bwd():
+    torch.manual_seed(1)
+    T, Hq, Hk, D = 130, 12, 2, 128
+    q = rand_bf16(T, Hq, D)
+    k = rand_bf16(T, Hk, D)
+    cos_t, sin_t = ops.build_rope_tables(4096, D, 10000.0, DEV)
+    pos = torch.randint(0, 4096, (T,), device=DEV, dtype=torch.int32)
+
+    q_out, k_out = ops.rope(q, k, cos_t, sin_t, pos)
+    q_ref, k_ref = ref.rope_ref(q, k, cos_t, sin_t, pos)
+    assert (q_out.float() - q_ref.float()).abs().max() < 2e-2
+    assert (k_out.float() - k_ref.float()).abs().max() < 2e-2
+
+    # backward = inverse rotation: rope_bwd(rope_fwd(g)) == g
+    qg = q.clone().requires_grad_(True)
+    kg = k.clone().requires_grad_(True)
+    qo, ko = ops.rope(qg, kg, cos_t, sin_t, pos)
+    (qo.float().sum() + ko.float().sum()).backward()
+    # reference grad via autograd on fp32
+    qf = q.float().clone().requires_grad_(True)
+    kf = k.float().clone().requires_grad_(True)
+    half = D // 2
+    c = cos_t[pos.long()].unsqueeze(1)
+    s = sin_t[pos.long()].unsqueeze(1)
+    for t, tf in ((qf, qf), (kf, kf)):
+        pass
+    qr = torch.cat([qf[..., :half] * c - qf[..., half:] * s, qf[..., half:] * c + qf[..., :half] * s], -1)
+    kr = torch.cat([kf[..., :half] * c - kf[..., half:] * s, kf[..., half:] * c + kf[..., :half] * s], -1)
+    (qr.sum() + kr.sum()).backward()
+    assert (qg.grad.float() - qf.grad).abs().max() < 2e-2
+    assert (kg.grad.float() - kf.grad).abs().max() < 2e-2
+
+
+@requires_gpu
+def test_swiglu_fwd_bwd():
+    torch.manual_seed(2)
+    gu = rand_bf16(123, 2 * 896)
+    out = ops.swiglu(gu)
+    out_ref = ref.swiglu_ref(gu)
+    assert (out.float() - out_ref.float()).abs().max() < 2e-2
+
+    g1 = gu.clone().requires_grad_(True)
+    o = ops.swiglu(g1)
+    gr = torch.randn_like(o, dtype=torch.float32).to(torch.bfloat16)
+    o.backward(gr)
+    g2 = gu.float().clone().requires_grad_(True)
+    gg, uu = g2.chunk(2, -1)
+    (torch.nn.functional.silu(gg) * uu).backward(gr.float())
+    assert (g1.grad.float() - g2.grad).abs().max() < 5e-2
+
+
+@requires_gpu
+def test_chunked_logprob_vs_ref():
+    torch.manual_seed(3)
+    T, H, V = 64, 512, 9984  # V multiple of 8, chunked at 4096
+    hidden = rand_bf16(T, H, scale=0.3)
+    w = rand_bf16(V, H, scale=0.3)
+    targets = torch.randint(0, V, (T,), device=DEV, dtype=torch.int64)
+
+    h1 = hidden.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    lp, ent = ops.chunked_logprob(h1, w1, targets, chunk=4096)
+    lp_ref, ent_ref = ref.logprob_entropy_ref(hidden, w, targets)
+    assert (lp - lp_ref).abs().max() < 0.05, (lp - lp_ref).abs().max()
+    assert (ent - ent_ref).abs().max() < 0.05
+
+    # backward vs fp32 autograd
+    dlp = torch.randn(T, device=DEV)
+    lp.backward(dlp)
+    h2 = hidden.float().clone().requires_grad_(True)
+    w2 = w.float().clone().requires_grad_(True)
+    logits = h2 @ w2.t()
+    lp2 = torch.log_softmax(logits, -1).gather(1, targets.unsqueeze(1)).squeeze(1)
+    lp2.backward(dlp)
+    assert (h1.grad.float() - h2.grad).abs().max() < 5e-2
+    # weight grad is [V, H]; compare scaled
+    assert (w1.grad.float() - w2.grad).abs().max() < 5e-2
+
+
+@requires_gpu
+def test_grpo_loss_vs_ref():
+    torch.manual_seed(4)
+    N = 4096
+    lp = torch.randn(N, device=DEV) * 0.3
+    old = lp + torch.randn(N, device=DEV) * 0.2
+    refp = lp + torch.randn(N, device=DEV) * 0.1
+    adv = torch.randn(N, device=DEV)
+    w = torch.rand(N, device=DEV) + 0.5
+
+    lp1 = lp.clone().requires_grad_(True)
+    loss, clipped = ops.grpo_loss_per_token(lp1, old, refp, adv, w, eps_lo=0.2, eps_hi=0.28, kl_beta=0.04)
+    loss_ref, clipped_ref = ref.grpo_loss_ref(lp, old, refp, adv, w, 0.2, 0.28, 0.04)
+    assert (loss - loss_ref).abs().max() < 1e-4
+    assert (clipped - clipped_ref).abs().max() < 1e-6
+
+    loss.sum().backward()
+    lp2 = lp.clone().requires_grad_(True)
+    l2, _ = ref.grpo_loss_ref(lp2, old, refp, adv, w, 0.2, 0.28, 0.04)
+    l2.sum().backward()
+    assert (lp1.grad - lp2.grad).abs().max() < 1e-4
+
+
+@requires_gpu
+def test_adamw_vs_ref():
+    torch.manual_seed(5)
+    N = 100003
+    master = torch.randn(N, device=DEV)
+    m = torch.randn(N, device=DEV).abs() * 0.01
+    v = torch.randn(N, device=DEV).abs() * 0.001
+    grad = rand_bf16(N, scale=0.1)
+    param = master.to(torch.bfloat16)
+
+    master_ref, m_ref, v_ref = ref.adamw_ref(
+        grad, master.clone(), m.clone(), v.clone(),
+        lr=1e-3, beta1=0.9, beta2=0.95, eps=1e-8, weight_decay=0.01, step=7,
+        grad_clip=1.0, grad_scale=0.5)
+
+    gnorm = ops.grad_sq_sum(grad, 0.5)
+    # check the squared-norm reduction itself
+    gn_ref = (grad.float() * 0.5).pow(2).sum()
+    assert torch.allclose(gnorm[0], gn_ref, rtol=1e-3)
+
+    ops.adamw_step(grad, master, m, v, param, gnorm, lr=1e-3, beta1=0.9, beta2=0.95,
+                   eps=1e-8, weight_decay=0.01, step=7, grad_clip=1.0, grad_scale=0.5)
+    assert (master - master_ref).abs().max() < 1e-5
+    assert (m - m_ref).abs().max() < 1e-5
+    assert (v - v_ref).abs().max() < 1e-6
+    assert (param.float() - master_ref).abs().max() < 1e-2
+
+
+def make_prefill_tiles(seqlens):
+    """Host-side q-tile list for flash_prefill."""
+    tile_seq_start, tile_row0, tile_seq_len = [], [], []
+    start = 0
+    for n in seqlens:
+        for r0 in range(0, n, 64):
+            tile_seq_start.append(start)
+            tile_row0.append(start + r0)
+            tile_seq_len.append(n)
+        start += n
+    t = lambda x: torch.tensor(x, device=DEV, dtype=torch.int32)
+    return t(tile_seq_start), t(tile_row0), t(tile_seq_len)
+
+
+@requires_gpu
+@pytest.mark.parametrize("seqlens", [[128], [65, 200, 64], [1, 333]])
+def test_flash_prefill_vs_ref(seqlens):
+    torch.manual_seed(6)
+    T = sum(seqlens)
+    Hq, Hk, D = 12, 2, 128
+    q = rand_bf16(T, Hq, D, scale=0.5)
+    k = rand_bf16(T, Hk, D, scale=0.5)
+    v = rand_bf16(T, Hk, D, scale=0.5)
+    scale = 1.0 / math.sqrt(D)
+    ts, tr, tl = make_prefill_tiles(seqlens)
+    o = ops.flash_prefill(q, k, v, ts, tr, tl, scale)
+    cu = [0]
+    for n in seqlens:
+        cu.append(cu[-1] + n)
+    o_ref = ref.attention_ref(q, k, v, cu, scale)
+    err = (o.float() - o_ref.float()).abs().max()
+    assert err < 3e-2, f"max err {err}"
+
+
+@requires_gpu
+@pytest.mark.parametrize("n_splits", [1, 4])
+def test_paged_decode_vs_ref(n_splits):
+    torch.manual_seed(7)
+    B, Hq, Hk, D = 5, 12, 2, 128
+    seq_lens = [1, 17, 256, 300, 64]
+    max_len = max(seq_lens)
+    n_pages_per_seq = (max_len + 15) // 16
+    total_pages = B * n_pages_per_seq + 1
+
+    k_pages = rand_bf16(total_pages, Hk, 16, D, scale=0.5)
+    v_pages = rand_bf16(total_pages, Hk, 16, D, scale=0.5)
+    q = rand_bf16(B, Hq, D, scale=0.5)
+    block_tables = torch.zeros(B, n_pages_per_seq, device=DEV, dtype=torch.int32)
+    for b in range(B):
+        for p in range((seq_lens[b] + 15) // 16):
+            block_tables[b, p] = 1 + b * n_pages_per_seq + p
+    seq_lens_t = torch.tensor(seq_lens, device=DEV, dtype=torch.int32)
+    scale = 1.0 / math.sqrt(D)
+
+    o = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens_t, scale, n_splits)
+
+    # reference: gather K/V per seq then dense attention of the last token
+    G = Hq // Hk
+    for b in range(B):
+        n = seq_lens[b]
+        ks = torch.zeros(n, Hk, D, device=DEV, dtype=torch.float32)
+        vs = torch.zeros(n, Hk, D, device=DEV, dtype=torch.float32)
+        for t in range(n):
+            page = int(block_tables[b, t // 16])
+            ks[t] = k_pages[page, :, t % 16].float()
+            vs[t] = v_pages[page, :, t % 16].float()
+        ks = ks.repeat_interleave(G, dim=1)
+        vs = vs.repeat_interleave(G, dim=1)
+        scores = torch.einsum("hd,khd->hk", q[b].float(), ks) * scale
+        p = torch.softmax(scores, dim=-1)
+        o_ref = torch.einsum("hk,khd->hd", p, vs)
+        err = (o[b].float() - o_ref).abs().max()
+        assert err < 3e-2, f"seq {b}: max err {err}"
+
+
+@requires_gpu
+def test_reshape_and_cache():
+    torch.manual_seed(8)
+    T, Hk, D = 40, 2, 128
+    k = rand_bf16(T, Hk, D)
+    v = rand_bf16(T, Hk, D)
+    k_pages = torch.zeros(8, Hk, 16, D, device=DEV, dtype=torch.bfloat16)
+    v_pages = torch.zeros(8, Hk, 16, D, device=DEV, dtype=torch.bfloat16)
+    slots = torch.arange(17, 17 + T, device=DEV, dtype=torch.int32)
+    ops.reshape_and_cache(k, v, k_pages, v_pages, slots)
+    for t in range(T):
+        slot = 17 + t
+        assert torch.equal(k_pages[slot // 16, :, slot % 16], k[t])
+        assert torch.equal(v_pages[slot // 16, :, slot % 16], v[t])
+
+
+@requires_gpu
+def test_sample_logprob_distribution():
+    torch.manual_seed(9)
+    # 4-token vocab padded to 8 with -inf; check empirical freq vs softmax
+    B, V = 4096, 8
+    base = torch.tensor([1.0, 0.0, -1.0, 2.0, -1e30, -1e30, -1e30, -1e30], device=DEV)
+    logits = base.expand(B, V).contiguous().to(torch.bfloat16)
+    tokens, logprobs = ops.sample_logprob(logits, 1.0, seed=123, step=0)
+    probs = torch.softmax(base[:4].float(), 0)
+    counts = torch.bincount(tokens.long(), minlength=V)[:4].float() / B
+    assert (counts - probs.cpu().to(counts.device)).abs().max() < 0.04, counts
+    # logprob correctness for sampled tokens
+    ref_lp = torch.log_softmax(logits.float(), -1)
+    expect = ref_lp.gather(1, tokens.long().unsqueeze(1)).squeeze(1)
+    assert (logprobs - expect).abs().max() < 2e-2
+
+
+@requires_gpu
+def test_sample_logprob_greedy_and_temperature():
+    torch.manual_seed(10)
+    B, V = 16, 1024
+    logits = rand_bf16(B, V, scale=2.0)
+    tokens, logprobs = ops.sample_logprob(logits, 0.0, seed=1, step=0)
+    assert torch.equal(tokens.long(), logits.float().argmax(-1))
+    # temperature changes the distribution's sharpness; just check logprob math
+    t2, lp2 = ops.sample_logprob(logits, 0.7, seed=2, step=5)
+    ref_lp = torch.log_softmax(logits.float() / 0.7, -1)
+    expect = ref_lp.gather(1, t2.long().unsqueeze(1)).squeeze(1)
+    assert (lp2 - expect).abs().max() < 3e-2
+
+
+@requires_gpu
+def test_gather_logprob():
+    torch.manual_seed(11)
+    T, V = 33, 2048
+    logits = rand_bf16(T, V)
+    toks = torch.randint(0, V, (T,), device=DEV, dtype=torch.int32)
+    lp = ops.gather_logprob(logits, toks, 1.0)
+    expect = torch.log_softmax(logits.float(), -1).gather(1, toks.long().unsqueeze(1)).squeeze(1)
+    assert (lp - expect).abs().max() < 2e-2
