@@ -1,0 +1,343 @@
+"""Continuous-batching LLM engine over the paged KV cache
+(K1-K6 in SURVEY.md §2.E — the vLLM replacement behind the gateway).
+
+Design (MI355X-first):
+* One engine per GPU; model weights are SHARED TENSORS with the trainer's
+  actor on the same device, so weight sync in colocated mode is free
+  (288 GB HBM: actor + ref + KV co-resident; no sleep/offload gymnastics).
+* step() = one scheduler iteration: admit waiting requests into a prefill
+  batch bounded by max_num_batched_tokens (reference knob yaml:86), else
+  one decode step over all running sequences (batch cap max_num_seqs).
+* Per-token logprob of the sampled token captured by the fused
+  sample_logprob kernel, 1:1 with completion ids (types.py:167-168).
+* abort/resume supported for partial rollout (async_agent_loop.py:11-14).
+"""
+
+from __future__ import annotations
+
+import math
+import time
+from dataclasses import dataclass, field
+from enum import Enum
+from typing import Any, Callable
+
+import torch
+
+from rllm_amd import ops
+from rllm_amd.engine.inference.kv_cache import PAGE_SIZE, KVCache
+from rllm_amd.models.qwen import QwenModel, make_prefill_tiles
+
+
+@dataclass
+class SamplingParams:
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1
+    max_tokens: int = 512
+    stop_token_ids: list[int] = field(default_factory=list)
+    logprobs: bool = True
+    seed: int = 0
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "SamplingParams":
+        return cls(
+            temperature=float(d.get("temperature", 1.0)),
+            top_p=float(d.get("top_p", 1.0)),
+            top_k=int(d.get("top_k", -1)),
+            max_tokens=int(d.get("max_tokens", d.get("max_completion_tokens", 512))),
+            stop_token_ids=list(d.get("stop_token_ids", [])),
+            logprobs=bool(d.get("logprobs", True)),
+            seed=int(d.get("seed", 0)),
+        )
+
+
+class SeqState(Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    FINISHED = "finished"
+    ABORTED = "aborted"
+
+
+class Sequence:
+    __slots__ = ("request_id", "prompt_ids", "output_ids", "logprobs", "params",
+                 "state", "pages", "finish_reason", "weight_version", "arrival_time",
+                 "first_token_time", "step_counter")
+
+    def __init__(self, request_id: str, prompt_ids: list[int], params: SamplingParams,
+                 weight_version: int = 0):
+        self.request_id = request_id
+        self.prompt_ids = list(prompt_ids)
+        self.output_ids: list[int] = []
+        self.logprobs: list[float] = []
+        self.params = params
+        self.state = SeqState.WAITING
+        self.pages: list[int] = []
+        self.finish_reason: str | None = None
+        self.weight_version = weight_version
+        self.arrival_time = time.monotonic()
+        self.first_token_time: float | None = None
+        self.step_counter = 0
+
+    @property
+    def total_len(self) -> int:
+        return len(self.prompt_ids) + len(self.output_ids)
+
+    @property
+    def last_token(self) -> int:
+        return self.output_ids[-1] if self.output_ids else self.prompt_ids[-1]
+
+
+@dataclass
+class RequestOutput:
+    request_id: str
+    prompt_ids: list[int]
+    token_ids: list[int]
+    logprobs: list[float]
+    finish_reason: str
+    weight_version: int
+    metrics: dict[str, float] = field(default_factory=dict)
+
+
+class LLMEngine:
+    """Synchronous engine core. Drive with step(); or use generate() for a
+    blocking batch. The async server wraps this in a background thread."""
+
+    def __init__(self, model: QwenModel, kv_cache: KVCache | None = None,
+                 max_num_seqs: int = 1024, max_num_batched_tokens: int = 8192,
+                 kv_budget_bytes: int | None = None, eos_token_id: int | None = None,
+                 seed: int = 0):
+        self.model = model
+        self.cfg = model.cfg
+        self.device = next(model.parameters()).device
+        if kv_cache is None:
+            if kv_budget_bytes is None:
+                free, _total = torch.cuda.mem_get_info(self.device)
+                kv_budget_bytes = int(free * 0.6)
+            kv_cache = KVCache.from_memory_budget(
+                self.cfg.num_layers, self.cfg.num_kv_heads, self.cfg.head_dim,
+                kv_budget_bytes, device=str(self.device))
+        self.kv = kv_cache
+        self.max_num_seqs = max_num_seqs
+        self.max_num_batched_tokens = max_num_batched_tokens
+        self.eos_token_id = eos_token_id
+        self.seed = seed
+        self.weight_version = 0
+        self._paused = False
+
+        self.waiting: list[Sequence] = []
+        self.running: list[Sequence] = []
+        self.finished: dict[str, Sequence] = {}
+        self._sample_step = 0
+        self._block_tables_cap = 0
+
+    # ------------------------------------------------------------------
+    # Request lifecycle
+    # ------------------------------------------------------------------
+    def add_request(self, request_id: str, prompt_ids: list[int], params: SamplingParams) -> None:
+        seq = Sequence(request_id, prompt_ids, params, weight_version=self.weight_version)
+        self.waiting.append(seq)
+
+    def abort(self, request_id: str) -> None:
+        for lst in (self.waiting, self.running):
+            for seq in lst:
+                if seq.request_id == request_id:
+                    lst.remove(seq)
+                    self._finish(seq, "abort")
+                    return
+
+    def has_unfinished(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def pause(self):
+        """Stop admitting new prefills (weight-sync drain protocol)."""
+        self._paused = True
+
+    def resume(self):
+        self._paused = False
+
+    def pop_finished(self) -> list[RequestOutput]:
+        outs = []
+        for seq in self.finished.values():
+            outs.append(RequestOutput(
+                request_id=seq.request_id,
+                prompt_ids=seq.prompt_ids,
+                token_ids=seq.output_ids,
+                logprobs=seq.logprobs,
+                finish_reason=seq.finish_reason or "stop",
+                weight_version=seq.weight_version,
+                metrics={"ttft_s": (seq.first_token_time - seq.arrival_time) if seq.first_token_time else 0.0},
+            ))
+        self.finished.clear()
+        return outs
+
+    # ------------------------------------------------------------------
+    # Scheduler step
+    # ------------------------------------------------------------------
+    def step(self) -> int:
+        """One iteration. Returns number of tokens processed."""
+        batch = self._schedule_prefill()
+        if batch:
+            return self._run_prefill(batch)
+        if self.running:
+            return self._run_decode()
+        return 0
+
+    def _schedule_prefill(self) -> list[Sequence]:
+        if self._paused or not self.waiting:
+            return []
+        batch: list[Sequence] = []
+        tokens = 0
+        while self.waiting and len(self.running) + len(batch) < self.max_num_seqs:
+            seq = self.waiting[0]
+            n = len(seq.prompt_ids)
+            if batch and tokens + n > self.max_num_batched_tokens:
+                break
+            need = KVCache.pages_needed(n + 1)
+            if need > self.kv.num_free_pages:
+                break
+            seq.pages = self.kv.alloc(need)
+            batch.append(self.waiting.pop(0))
+            tokens += n
+        return batch
+
+    def _slot(self, seq: Sequence, pos: int) -> int:
+        return seq.pages[pos // PAGE_SIZE] * PAGE_SIZE + pos % PAGE_SIZE
+
+    def _run_prefill(self, batch: list[Sequence]) -> int:
+        device = self.device
+        input_ids, positions, slot_mapping, seqlens, last_rows = [], [], [], [], []
+        row = 0
+        for seq in batch:
+            n = len(seq.prompt_ids)
+            input_ids.extend(seq.prompt_ids)
+            positions.extend(range(n))
+            slot_mapping.extend(self._slot(seq, p) for p in range(n))
+            seqlens.append(n)
+            row += n
+            last_rows.append(row - 1)
+
+        ids_t = torch.tensor(input_ids, device=device, dtype=torch.long)
+        pos_t = torch.tensor(positions, device=device, dtype=torch.int32)
+        slots_t = torch.tensor(slot_mapping, device=device, dtype=torch.int32)
+        tiles = make_prefill_tiles(seqlens, device)
+
+        hidden = self.model.forward_prefill(ids_t, pos_t, tiles, self.kv, slots_t)
+        last_hidden = hidden[torch.tensor(last_rows, device=device, dtype=torch.long)]
+        self._sample_and_append(batch, last_hidden)
+        for seq in batch:
+            if seq.state != SeqState.FINISHED:
+                seq.state = SeqState.RUNNING
+                self.running.append(seq)
+            if seq.first_token_time is None:
+                seq.first_token_time = time.monotonic()
+        return len(input_ids)
+
+    def _run_decode(self) -> int:
+        device = self.device
+        batch = self.running
+        B = len(batch)
+        input_ids, positions, slot_mapping, seq_lens = [], [], [], []
+        for seq in batch:
+            pos = seq.total_len  # position of the NEW token
+            # grow page table if needed
+            if KVCache.pages_needed(pos + 1) > len(seq.pages):
+                seq.pages.extend(self.kv.alloc(1))
+            input_ids.append(seq.last_token)
+            positions.append(pos)
+            slot_mapping.append(self._slot(seq, pos))
+            seq_lens.append(pos + 1)
+
+        max_pages = max(len(s.pages) for s in batch)
+        block_tables = torch.zeros(B, max_pages, device=device, dtype=torch.int32)
+        for i, seq in enumerate(batch):
+            block_tables[i, : len(seq.pages)] = torch.tensor(seq.pages, dtype=torch.int32)
+
+        ids_t = torch.tensor(input_ids, device=device, dtype=torch.long)
+        pos_t = torch.tensor(positions, device=device, dtype=torch.int32)
+        slots_t = torch.tensor(slot_mapping, device=device, dtype=torch.int32)
+        lens_t = torch.tensor(seq_lens, device=device, dtype=torch.int32)
+
+        hidden = self.model.forward_decode(ids_t, pos_t, self.kv, slots_t, block_tables, lens_t)
+        self._sample_and_append(batch, hidden)
+        self.running = [s for s in batch if s.state == SeqState.RUNNING]
+        return B
+
+    # ------------------------------------------------------------------
+    def _sample_and_append(self, batch: list[Sequence], hidden: torch.Tensor) -> None:
+        logits = self.model.logits(hidden)  # [B, V] bf16
+        # group rows by (temperature, top_k, top_p); default path is one fused call
+        temp = batch[0].params.temperature
+        uniform = all(s.params.temperature == temp and s.params.top_k <= 0 and s.params.top_p >= 1.0
+                      for s in batch)
+        self._sample_step += 1
+        if uniform:
+            tokens, logprobs = ops.sample_logprob(logits.contiguous(), temp, self.seed, self._sample_step)
+            tokens = tokens.tolist()
+            logprobs = logprobs.tolist()
+        else:
+            tokens, logprobs = [], []
+            for i, seq in enumerate(batch):
+                row = logits[i : i + 1]
+                p = seq.params
+                if p.top_k > 0 or p.top_p < 1.0:
+                    row = _mask_top_k_top_p(row.float(), p.top_k, p.top_p, p.temperature).to(torch.bfloat16)
+                t, lp = ops.sample_logprob(row.contiguous(), p.temperature, self.seed + i, self._sample_step)
+                tokens.append(int(t[0]))
+                logprobs.append(float(lp[0]))
+
+        for seq, tok, lp in zip(batch, tokens, logprobs):
+            seq.output_ids.append(int(tok))
+            seq.logprobs.append(float(lp))
+            self._maybe_finish(seq)
+
+    def _maybe_finish(self, seq: Sequence) -> None:
+        tok = seq.output_ids[-1]
+        stop_ids = set(seq.params.stop_token_ids)
+        if self.eos_token_id is not None:
+            stop_ids.add(self.eos_token_id)
+        if tok in stop_ids:
+            self._finish(seq, "stop")
+        elif len(seq.output_ids) >= seq.params.max_tokens:
+            self._finish(seq, "length")
+
+    def _finish(self, seq: Sequence, reason: str) -> None:
+        seq.state = SeqState.FINISHED
+        seq.finish_reason = reason
+        if seq.pages:
+            self.kv.free(seq.pages)
+            seq.pages = []
+        self.finished[seq.request_id] = seq
+
+    # ------------------------------------------------------------------
+    def generate(self, prompts: list[list[int]], params: SamplingParams | list[SamplingParams],
+                 on_step: Callable[[], None] | None = None) -> list[RequestOutput]:
+        """Blocking batch generation (bench / tests)."""
+        if isinstance(params, SamplingParams):
+            params = [params] * len(prompts)
+        for i, (p, sp) in enumerate(zip(prompts, params)):
+            self.add_request(f"gen-{i}", p, sp)
+        while self.has_unfinished():
+            self.step()
+            if on_step is not None:
+                on_step()
+        outs = {o.request_id: o for o in self.pop_finished()}
+        return [outs[f"gen-{i}"] for i in range(len(prompts))]
+
+
+def _mask_top_k_top_p(logits_f32: torch.Tensor, top_k: int, top_p: float, temperature: float) -> torch.Tensor:
+    """Pre-mask logits (-inf) for top-k / nucleus sampling; the fused kernel
+    then samples the masked distribution exactly. Rare path — the training
+    defaults are temp 1.0 / top_p 1.0 (reference base.yaml:49-60)."""
+    z = logits_f32 / max(temperature, 1e-6)
+    if top_k > 0 and top_k < z.shape[-1]:
+        kth = torch.topk(z, top_k, dim=-1).values[..., -1:]
+        logits_f32 = logits_f32.masked_fill(z < kth, float("-inf"))
+        z = z.masked_fill(z < kth, float("-inf"))
+    if top_p < 1.0:
+        sorted_z, idx = torch.sort(z, descending=True, dim=-1)
+        probs = torch.softmax(sorted_z, dim=-1)
+        cum = probs.cumsum(-1)
+        keep_sorted = cum - probs < top_p  # always keep the top token
+        keep = torch.zeros_like(keep_sorted).scatter(-1, idx, keep_sorted)
+        logits_f32 = logits_f32.masked_fill(~keep, float("-inf"))
+    return logits_f32

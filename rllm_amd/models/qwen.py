@@ -1,0 +1,258 @@
+"""Qwen2-family decoder, MI355X-native.
+
+Three forward paths over one set of bf16 weights:
+
+* ``forward_train`` — packed varlen, autograd-enabled. Elementwise/norm ops
+  are the in-repo HIP kernels (rmsnorm / rope / swiglu autograd wrappers);
+  projections are hipBLASLt GEMMs (torch.matmul); attention is a chunked
+  bf16-matmul + fp32-softmax composition (the update-path attention; the
+  hand-written flash kernels serve the rollout path — SURVEY.md §2.E K5/K7
+  vs K1/K2). The LM head goes through ops.chunked_logprob so the [T, vocab]
+  logits matrix is never materialized.
+* ``forward_prefill`` — no-grad rollout prefill: flash_prefill HIP kernel +
+  reshape_and_cache into KV pages.
+* ``forward_decode`` — no-grad one-token step over the paged KV cache.
+
+Weight layout is fused (qkv_proj, gate_up_proj) for GEMM efficiency;
+``load_hf_state_dict`` maps HuggingFace Qwen2 checkpoints onto it.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from rllm_amd import ops
+from rllm_amd.models.config import ModelConfig
+
+
+def _linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None = None) -> torch.Tensor:
+    return torch.nn.functional.linear(x, w, b)
+
+
+def train_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    cu_seqlens: list[int], scale: float, q_chunk: int = 1024) -> torch.Tensor:
+    """Causal GQA attention for the update path: bf16 GEMMs (hipBLASLt) with
+    fp32 softmax, chunked over query blocks to bound the score matrix.
+
+    q [T, Hq, D], k/v [T, Hk, D] packed varlen; returns [T, Hq, D] bf16.
+    """
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    G = Hq // Hk
+    out = torch.empty_like(q)
+    for b in range(len(cu_seqlens) - 1):
+        s0, s1 = cu_seqlens[b], cu_seqlens[b + 1]
+        n = s1 - s0
+        if n == 0:
+            continue
+        qs = q[s0:s1].permute(1, 0, 2)                       # [Hq, n, D]
+        ks = k[s0:s1].permute(1, 0, 2)                       # [Hk, n, D]
+        vs = v[s0:s1].permute(1, 0, 2)
+        if G > 1:
+            ks = ks.repeat_interleave(G, dim=0)
+            vs = vs.repeat_interleave(G, dim=0)
+        rows = []
+        for c0 in range(0, n, q_chunk):
+            c1 = min(n, c0 + q_chunk)
+            scores = torch.matmul(qs[:, c0:c1], ks.transpose(1, 2)).float() * scale  # [Hq, c, n]
+            mask = torch.arange(n, device=q.device).unsqueeze(0) > torch.arange(c0, c1, device=q.device).unsqueeze(1)
+            scores.masked_fill_(mask.unsqueeze(0), float("-inf"))
+            p = torch.softmax(scores, dim=-1).to(q.dtype)
+            rows.append(torch.matmul(p, vs))                 # [Hq, c, D]
+        out[s0:s1] = torch.cat(rows, dim=1).permute(1, 0, 2)
+    return out
+
+
+class QwenLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        H = cfg.hidden_size
+        self.cfg = cfg
+        qkv_out = cfg.q_size + 2 * cfg.kv_size
+        self.input_layernorm = nn.Parameter(torch.ones(H, dtype=torch.bfloat16))
+        self.qkv_proj = nn.Parameter(torch.empty(qkv_out, H, dtype=torch.bfloat16))
+        self.qkv_bias = nn.Parameter(torch.zeros(qkv_out, dtype=torch.bfloat16)) if cfg.qkv_bias else None
+        self.o_proj = nn.Parameter(torch.empty(H, cfg.q_size, dtype=torch.bfloat16))
+        self.post_attention_layernorm = nn.Parameter(torch.ones(H, dtype=torch.bfloat16))
+        self.gate_up_proj = nn.Parameter(torch.empty(2 * cfg.intermediate_size, H, dtype=torch.bfloat16))
+        self.down_proj = nn.Parameter(torch.empty(H, cfg.intermediate_size, dtype=torch.bfloat16))
+
+    def _qkv(self, hidden: torch.Tensor, positions: torch.Tensor, cos_t, sin_t):
+        cfg = self.cfg
+        T = hidden.shape[0]
+        x = ops.rmsnorm(hidden, self.input_layernorm, cfg.rms_eps)
+        qkv = _linear(x, self.qkv_proj, self.qkv_bias)
+        q, k, v = qkv.split([cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1)
+        q = q.view(T, cfg.num_heads, cfg.head_dim)
+        k = k.view(T, cfg.num_kv_heads, cfg.head_dim)
+        v = v.view(T, cfg.num_kv_heads, cfg.head_dim).contiguous()
+        q, k = ops.rope(q, k, cos_t, sin_t, positions)
+        return q, k, v
+
+    def _finish(self, hidden: torch.Tensor, attn_out: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        T = hidden.shape[0]
+        hidden = hidden + _linear(attn_out.reshape(T, cfg.q_size), self.o_proj)
+        x = ops.rmsnorm(hidden, self.post_attention_layernorm, cfg.rms_eps)
+        mlp = _linear(ops.swiglu(_linear(x, self.gate_up_proj)), self.down_proj)
+        return hidden + mlp
+
+    # -- training path -------------------------------------------------------
+    def forward_train(self, hidden, positions, cu_seqlens, cos_t, sin_t):
+        q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
+        attn = train_attention(q, k, v, cu_seqlens, 1.0 / math.sqrt(self.cfg.head_dim))
+        return self._finish(hidden, attn)
+
+    # -- rollout paths (no grad) ----------------------------------------------
+    @torch.no_grad()
+    def forward_prefill(self, hidden, positions, cos_t, sin_t, tiles, kv_cache, slot_mapping, layer_idx):
+        q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
+        k_pages, v_pages = kv_cache[layer_idx]
+        ops.reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
+        attn = ops.flash_prefill(q, k, v, tiles[0], tiles[1], tiles[2],
+                                 1.0 / math.sqrt(self.cfg.head_dim))
+        return self._finish(hidden, attn)
+
+    @torch.no_grad()
+    def forward_decode(self, hidden, positions, cos_t, sin_t, kv_cache, slot_mapping,
+                       block_tables, seq_lens, layer_idx):
+        q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
+        k_pages, v_pages = kv_cache[layer_idx]
+        ops.reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
+        attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
+                                1.0 / math.sqrt(self.cfg.head_dim))
+        return self._finish(hidden, attn)
+
+
+class QwenModel(nn.Module):
+    def __init__(self, cfg: ModelConfig, device: str = "cuda", max_positions: int = 65536):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Parameter(torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+        self.layers = nn.ModuleList([QwenLayer(cfg) for _ in range(cfg.num_layers)])
+        self.norm = nn.Parameter(torch.ones(cfg.hidden_size, dtype=torch.bfloat16))
+        if cfg.tie_word_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = nn.Parameter(torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+        self.gradient_checkpointing = False
+        self._device = device
+        self.to(device)
+        cos_t, sin_t = ops.build_rope_tables(max_positions, cfg.head_dim, cfg.rope_theta, device) \
+            if device != "meta" else (None, None)
+        self.cos_t, self.sin_t = cos_t, sin_t
+
+    @property
+    def lm_weight(self) -> torch.Tensor:
+        return self.embed_tokens if self.lm_head is None else self.lm_head
+
+    def init_random(self, seed: int = 0):
+        """Random init at HF-like scales (std 0.02)."""
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        for name, p in self.named_parameters():
+            if p.dim() == 2:
+                init = torch.randn(p.shape, generator=gen, dtype=torch.float32) * 0.02
+                with torch.no_grad():
+                    p.copy_(init.to(p.dtype))
+            elif "bias" in name:
+                with torch.no_grad():
+                    p.zero_()
+            else:  # norms
+                with torch.no_grad():
+                    p.fill_(1.0)
+        return self
+
+    # -- training -------------------------------------------------------------
+    def forward_train(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                      cu_seqlens: list[int]) -> torch.Tensor:
+        """Packed varlen forward -> final hidden states [T, H] (after norm)."""
+        hidden = self.embed_tokens[input_ids]
+        for layer in self.layers:
+            if self.gradient_checkpointing and torch.is_grad_enabled():
+                hidden = torch.utils.checkpoint.checkpoint(
+                    layer.forward_train, hidden, positions, cu_seqlens,
+                    self.cos_t, self.sin_t, use_reentrant=False)
+            else:
+                hidden = layer.forward_train(hidden, positions, cu_seqlens, self.cos_t, self.sin_t)
+        return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
+
+    def logprobs_for_tokens(self, input_ids, positions, cu_seqlens, targets,
+                            want_entropy: bool = True, temperature: float = 1.0):
+        """Token logprob (+ entropy) for `targets` aligned with input rows.
+        The caller masks out rows it does not need (prompt/observation rows
+        get target -1 ... they still cost a gather, so callers typically
+        index-select response rows first)."""
+        hidden = self.forward_train(input_ids, positions, cu_seqlens)
+        return ops.chunked_logprob(hidden, self.lm_weight, targets,
+                                   temperature=temperature, want_entropy=want_entropy)
+
+    # -- rollout --------------------------------------------------------------
+    @torch.no_grad()
+    def forward_prefill(self, input_ids, positions, tiles, kv_cache, slot_mapping):
+        hidden = self.embed_tokens[input_ids]
+        for i, layer in enumerate(self.layers):
+            hidden = layer.forward_prefill(hidden, positions, self.cos_t, self.sin_t,
+                                           tiles, kv_cache, slot_mapping, i)
+        return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
+
+    @torch.no_grad()
+    def forward_decode(self, input_ids, positions, kv_cache, slot_mapping, block_tables, seq_lens):
+        hidden = self.embed_tokens[input_ids]
+        for i, layer in enumerate(self.layers):
+            hidden = layer.forward_decode(hidden, positions, self.cos_t, self.sin_t,
+                                          kv_cache, slot_mapping, block_tables, seq_lens, i)
+        return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
+
+    @torch.no_grad()
+    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return torch.matmul(hidden, self.lm_weight.t())
+
+    # -- HF checkpoint mapping --------------------------------------------------
+    @torch.no_grad()
+    def load_hf_state_dict(self, sd: dict[str, torch.Tensor]):
+        """Map a HuggingFace Qwen2 state dict onto the fused layout."""
+        cfg = self.cfg
+
+        def cp(dst: torch.Tensor, key: str):
+            dst.copy_(sd[key].to(dst.dtype))
+
+        cp(self.embed_tokens, "model.embed_tokens.weight")
+        cp(self.norm, "model.norm.weight")
+        if self.lm_head is not None:
+            cp(self.lm_head, "lm_head.weight")
+        for i, layer in enumerate(self.layers):
+            p = f"model.layers.{i}."
+            cp(layer.input_layernorm, p + "input_layernorm.weight")
+            cp(layer.post_attention_layernorm, p + "post_attention_layernorm.weight")
+            q = sd[p + "self_attn.q_proj.weight"]
+            k = sd[p + "self_attn.k_proj.weight"]
+            v = sd[p + "self_attn.v_proj.weight"]
+            layer.qkv_proj.copy_(torch.cat([q, k, v], dim=0).to(torch.bfloat16))
+            if layer.qkv_bias is not None:
+                qb = sd[p + "self_attn.q_proj.bias"]
+                kb = sd[p + "self_attn.k_proj.bias"]
+                vb = sd[p + "self_attn.v_proj.bias"]
+                layer.qkv_bias.copy_(torch.cat([qb, kb, vb], dim=0).to(torch.bfloat16))
+            cp(layer.o_proj, p + "self_attn.o_proj.weight")
+            g = sd[p + "mlp.gate_proj.weight"]
+            u = sd[p + "mlp.up_proj.weight"]
+            layer.gate_up_proj.copy_(torch.cat([g, u], dim=0).to(torch.bfloat16))
+            cp(layer.down_proj, p + "mlp.down_proj.weight")
+        return self
+
+
+def make_prefill_tiles(seqlens: list[int], device, tile: int = 64):
+    """Host-side q-tile table for flash_prefill over packed varlen batches."""
+    tile_seq_start, tile_row0, tile_seq_len = [], [], []
+    start = 0
+    for n in seqlens:
+        for r0 in range(0, n, tile):
+            tile_seq_start.append(start)
+            tile_row0.append(start + r0)
+            tile_seq_len.append(n)
+        start += n
+    mk = lambda x: torch.tensor(x, device=device, dtype=torch.int32)
+    return mk(tile_seq_start), mk(tile_row0), mk(tile_seq_len)

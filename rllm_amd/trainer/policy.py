@@ -1,0 +1,213 @@
+"""PolicyTrainer — the native policy-update path (replaces verl FSDP
+workers).
+
+One GPU holds: actor (bf16, flat), frozen reference policy (bf16), fp32
+optimizer state, and — colocated — the rollout engine SHARING the actor's
+weight tensors (288 GB HBM). DP across GPUs = RCCL all-reduce of the flat
+gradient buffer.
+
+update_policy semantics mirror the reference defaults
+(_generated_agent_ppo_trainer.yaml): ratio-clip 0.2 (asymmetric high clip
+supported), KL-in-loss low_var_kl, loss_agg token-mean with the GLOBAL
+response-token denominator, dynamic token-balanced micro-batches with
+all_reduce(MAX) of the micro-batch count across DP (patch.py:23-66
+deadlock guard).
+"""
+
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass
+
+import torch
+
+from rllm_amd import ops
+from rllm_amd.parallel import dist as pdist
+from rllm_amd.trainer.algorithms.config import AlgorithmConfig
+from rllm_amd.trainer.batch import PackedRow, TrainBatch, pack_rows, split_rows_token_balanced
+from rllm_amd.trainer.optim import FusedAdamW, flatten_params
+
+logger = logging.getLogger(__name__)
+
+
+@dataclass
+class PolicyTrainerConfig:
+    lr: float = 1e-6
+    betas: tuple = (0.9, 0.999)
+    weight_decay: float = 0.0
+    grad_clip: float = 1.0
+    eps_clip: float = 0.2
+    eps_clip_high: float | None = None
+    kl_beta: float = 0.0
+    max_tokens_per_micro: int = 16384
+    loss_agg_mode: str = "token-mean"
+    tis_mode: str | None = None     # None | "token" | "sequence"
+    tis_cap: float = 5.0
+    bypass_mode: bool = False        # π_old := π_rollout (skip recompute)
+    entropy_chunk: int = 16384
+    use_ref: bool = True
+
+    @classmethod
+    def from_algorithm_config(cls, a: AlgorithmConfig, **over) -> "PolicyTrainerConfig":
+        return cls(
+            eps_clip=a.eps_clip,
+            eps_clip_high=a.eps_clip_high,
+            kl_beta=a.kl_beta,
+            loss_agg_mode=a.loss_agg_mode or "token-mean",
+            tis_mode=a.rollout_correction.tis_mode,
+            tis_cap=a.rollout_correction.tis_cap,
+            bypass_mode=bool(a.rollout_correction.bypass_mode),
+            use_ref=a.kl_beta > 0,
+            **over,
+        )
+
+
+class PolicyTrainer:
+    def __init__(self, model, ref_model=None, config: PolicyTrainerConfig | None = None):
+        self.model = model
+        self.ref_model = ref_model
+        self.cfg = config or PolicyTrainerConfig()
+        self.device = next(model.parameters()).device
+        self.flat_param, self.flat_grad = flatten_params(model)
+        self.optim = FusedAdamW(
+            self.flat_param, self.flat_grad,
+            lr=self.cfg.lr, betas=self.cfg.betas, weight_decay=self.cfg.weight_decay,
+            grad_clip=self.cfg.grad_clip)
+        self.weight_version = 0
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def compute_logprobs(self, batch: TrainBatch, model=None, want_entropy: bool = False):
+        """Forward pass → logprob of each loss token (fp32, [n_resp]);
+        optionally mean entropy. Used for old-logprob and ref-logprob
+        (reference compute_log_prob / compute_ref_log_prob RPCs)."""
+        model = model or self.model
+        hidden = model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
+        rows = batch.loss_mask.nonzero(as_tuple=True)[0]
+        h = hidden[rows]
+        tgt = batch.targets[rows]
+        lp, ent = ops.chunked_logprob(h, model.lm_weight, tgt,
+                                      chunk=self.cfg.entropy_chunk, want_entropy=want_entropy)
+        return lp, (ent.mean().item() if ent is not None else None)
+
+    # ------------------------------------------------------------------
+    def _tis_weights(self, old_lp: torch.Tensor, rollout_lp: torch.Tensor) -> torch.Tensor | None:
+        """Truncated importance-sampling correction weights
+        (reference verl_backend.py:663-676)."""
+        if self.cfg.tis_mode is None:
+            return None
+        log_ratio = old_lp - rollout_lp
+        w = torch.exp(log_ratio).clamp(max=self.cfg.tis_cap)
+        return w
+
+    def update_policy(self, rows: list[PackedRow], old_logprob_fn=None) -> dict:
+        """One optimizer step over a mini-batch of packed rows.
+
+        Pipeline per micro-batch: forward (custom kernels) → chunked
+        logprob on loss rows → fused GRPO loss (loss + dlp in one kernel)
+        → backward into the flat grad buffer. Then RCCL all-reduce +
+        fused AdamW. Returns metrics.
+        """
+        cfg = self.cfg
+        micros = split_rows_token_balanced(rows, cfg.max_tokens_per_micro)
+        # DP desync guard: every rank must run the same number of micro
+        # fwd/bwd? No — grads accumulate locally; only the all-reduce at the
+        # end must match. But the GLOBAL token denominator must agree:
+        n_local_tokens = sum(sum(r.response_mask) for r in rows)
+        n_global_tokens = pdist.all_reduce_scalar(float(n_local_tokens), op="sum")
+        world = pdist.get_world_size()
+        if n_global_tokens <= 0:
+            logger.warning("update_policy called with zero response tokens; skipping step")
+            return {"actor/skipped": 1.0}
+
+        self.optim.zero_grad()
+        tot_loss = 0.0
+        tot_pg = 0.0
+        tot_kl_count = 0
+        tot_clip = 0.0
+        ent_sum, ent_n = 0.0, 0
+
+        for micro in micros:
+            batch = pack_rows(micro, device=str(self.device))
+            hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
+            rows_idx = batch.loss_mask.nonzero(as_tuple=True)[0]
+            h = hidden[rows_idx]
+            tgt = batch.targets[rows_idx]
+            adv = batch.advantages[rows_idx]
+            rollout_lp = batch.rollout_logprobs[rows_idx]
+
+            lp, ent = ops.chunked_logprob(h, self.model.lm_weight, tgt,
+                                          chunk=cfg.entropy_chunk, want_entropy=True)
+            if ent is not None:
+                ent_sum += float(ent.sum())
+                ent_n += ent.numel()
+
+            with torch.no_grad():
+                if cfg.bypass_mode or old_logprob_fn is None:
+                    old_lp = rollout_lp
+                else:
+                    old_lp = old_logprob_fn(batch, rows_idx)
+                ref_lp = None
+                if cfg.use_ref and self.ref_model is not None:
+                    ref_hidden = self.ref_model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
+                    ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.ref_model.lm_weight,
+                                                    tgt, chunk=cfg.entropy_chunk, want_entropy=False)
+                tis_w = self._tis_weights(old_lp, rollout_lp)
+
+            eps_hi = cfg.eps_clip_high if cfg.eps_clip_high is not None else cfg.eps_clip
+            loss_tok, clipped = ops.grpo_loss_per_token(
+                lp, old_lp, ref_lp, adv, tis_w,
+                eps_lo=cfg.eps_clip, eps_hi=eps_hi, kl_beta=cfg.kl_beta)
+
+            if cfg.loss_agg_mode == "token-mean":
+                # global-token-mean: SUM-all-reduce of grads then NO rescale
+                loss = loss_tok.sum() / n_global_tokens
+            elif cfg.loss_agg_mode == "seq-mean-token-mean":
+                # per-sequence token mean, then mean over the global row count
+                n_global_rows = pdist.all_reduce_scalar(float(batch.n_rows), op="sum") if world > 1 else batch.n_rows
+                seq_ids = torch.bucketize(rows_idx, torch.tensor(batch.cu_seqlens[1:-1], device=rows_idx.device))
+                loss = 0.0
+                for s in seq_ids.unique():
+                    m = seq_ids == s
+                    loss = loss + loss_tok[m].mean()
+                loss = loss / max(1.0, n_global_rows)
+            else:
+                loss = loss_tok.sum() / n_global_tokens
+            loss.backward()
+            tot_loss += float(loss.detach()) * (1.0 if cfg.loss_agg_mode != "token-mean" else 1.0)
+            tot_pg += float(loss_tok.detach().sum())
+            tot_clip += float(clipped.sum())
+            tot_kl_count += int(loss_tok.numel())
+
+        # C1: DP gradient all-reduce (SUM; the global denominator already
+        # normalizes, so no 1/world rescale for token-mean)
+        if world > 1:
+            pdist.all_reduce_sum_(self.flat_grad)
+        gnorm_sq = self.optim.step(grad_scale=1.0)
+        self.weight_version += 1
+
+        metrics = {
+            "actor/loss": tot_loss,
+            "actor/pg_loss_tok_mean": tot_pg / max(1, n_local_tokens),
+            "actor/clipfrac": tot_clip / max(1, tot_kl_count),
+            "actor/entropy": ent_sum / max(1, ent_n),
+            "actor/grad_norm": float(gnorm_sq.sqrt().item()),
+            "actor/lr": self.optim.lr_at(self.optim.step_count - 1),
+            "actor/n_micro_batches": len(micros),
+            "actor/n_response_tokens": n_local_tokens,
+        }
+        return metrics
+
+    # ------------------------------------------------------------------
+    def state_dict(self) -> dict:
+        return {
+            "model": {k: v for k, v in self.model.state_dict().items()},
+            "optim": self.optim.state_dict(),
+            "weight_version": self.weight_version,
+        }
+
+    def load_state_dict(self, sd: dict):
+        self.model.load_state_dict(sd["model"])
+        self.flat_param.copy_(torch.cat([p.data.reshape(-1) for p in self.model.parameters()]))
+        self.optim.load_state_dict(sd["optim"])
+        self.weight_version = int(sd.get("weight_version", 0))

@@ -61,8 +61,9 @@ def test_rope_fwd_bwd():
 
     q_out, k_out = ops.rope(q, k, cos_t, sin_t, pos)
     q_ref, k_ref = ref.rope_ref(q, k, cos_t, sin_t, pos)
-    assert (q_out.float() - q_ref.float()).abs().max() < 2e-2
-    assert (k_out.float() - k_ref.float()).abs().max() < 2e-2
+    # fp32 fma-order differences can flip one bf16 ULP (0.03 at |x|~4)
+    assert (q_out.float() - q_ref.float()).abs().max() < 5e-2
+    assert (k_out.float() - k_ref.float()).abs().max() < 5e-2
 
     # backward = inverse rotation: rope_bwd(rope_fwd(g)) == g
     qg = q.clone().requires_grad_(True)
@@ -80,8 +81,8 @@ def test_rope_fwd_bwd():
     qr = torch.cat([qf[..., :half] * c - qf[..., half:] * s, qf[..., half:] * c + qf[..., :half] * s], -1)
     kr = torch.cat([kf[..., :half] * c - kf[..., half:] * s, kf[..., half:] * c + kf[..., :half] * s], -1)
     (qr.sum() + kr.sum()).backward()
-    assert (qg.grad.float() - qf.grad).abs().max() < 2e-2
-    assert (kg.grad.float() - kf.grad).abs().max() < 2e-2
+    assert (qg.grad.float() - qf.grad).abs().max() < 5e-2
+    assert (kg.grad.float() - kf.grad).abs().max() < 5e-2
 
 
 @requires_gpu
@@ -178,7 +179,9 @@ def test_adamw_vs_ref():
     assert (master - master_ref).abs().max() < 1e-5
     assert (m - m_ref).abs().max() < 1e-5
     assert (v - v_ref).abs().max() < 1e-6
-    assert (param.float() - master_ref).abs().max() < 1e-2
+    # param is the bf16 rounding of master: compare relatively (bf16 rel err <= 2^-8)
+    rel = ((param.float() - master_ref) / (master_ref.abs() + 1.0)).abs().max()
+    assert rel < 5e-3, rel
 
 
 def make_prefill_tiles(seqlens):
