@@ -109,8 +109,8 @@ class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos_tab, sin_tab, positions):
         C = require_ext()
-        q = q.contiguous().clone()
-        k = k.contiguous().clone()
+        q = q.clone(memory_format=torch.contiguous_format)
+        k = k.clone(memory_format=torch.contiguous_format)
         C.rope_inplace(q, k, cos_tab, sin_tab, positions, False)
         ctx.save_for_backward(cos_tab, sin_tab, positions)
         return q, k
@@ -126,12 +126,13 @@ class _RoPEFn(torch.autograd.Function):
 
 
 def rope(q, k, cos_tab, sin_tab, positions):
-    """Apply rotary embedding. q [T,Hq,D], k [T,Hk,D], positions [T] int32."""
+    """Apply rotary embedding (never aliases its inputs).
+    q [T,Hq,D], k [T,Hk,D], positions [T] int32."""
     if q.requires_grad or k.requires_grad:
         return _RoPEFn.apply(q, k, cos_tab, sin_tab, positions)
     C = require_ext()
-    q = q.contiguous()
-    k = k.contiguous()
+    q = q.clone(memory_format=torch.contiguous_format)
+    k = k.clone(memory_format=torch.contiguous_format)
     C.rope_inplace(q, k, cos_tab, sin_tab, positions, False)
     return q, k
 
@@ -170,6 +171,8 @@ class _ChunkedLogprobFn(torch.autograd.Function):
         lse = m + torch.log(s)
         logprob = tgt_logit - lse
         entropy = (lse - e / s) if want_entropy else None
+        if entropy is not None:
+            ctx.mark_non_differentiable(entropy)
         ctx.save_for_backward(hidden, lm_weight, targets_i32, lse)
         ctx.chunk = chunk
         ctx.inv_temp = inv_temp
