@@ -162,9 +162,6 @@ class UnifiedTrainer:
         episodes = await self.backend.generate_episodes(tasks, uids=uids)
         metrics["time/gen_s"] = time.monotonic() - t
         self.state.total_episodes += len(episodes)
-        if self.episode_logger:
-            self.episode_logger.log_episodes(episodes, mode="train",
-                                             step=self.state.global_step, epoch=self.state.epoch)
         metrics.update(_termination_metrics(episodes))
 
         # S2 — transform to trajectory groups
@@ -206,6 +203,10 @@ class UnifiedTrainer:
         self.state.total_response_tokens += n_resp
         metrics["batch/response_tokens"] = n_resp
         metrics["batch/num_episodes"] = len(episodes)
+        # episode logs carry the post-advantage training payloads
+        if self.episode_logger:
+            self.episode_logger.log_episodes(episodes, mode="train",
+                                             step=self.state.global_step, epoch=self.state.epoch)
         return metrics
 
     # ------------------------------------------------------------------

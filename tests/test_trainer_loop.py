@@ -1,0 +1,103 @@
+"""Full training-loop integration on CPU (BASELINE config 1 shape):
+UnifiedTrainer + CPUBackend + real gateway + @rollout/@evaluator.
+
+Covers: the 8-stage batch, GRPO advantages landing in the update,
+metrics, checkpoint save/resume, validation, rejection sampling skip.
+"""
+
+import sys
+from pathlib import Path
+
+import httpx
+import pytest
+
+sys.path.insert(0, str(Path(__file__).parent))
+
+import rllm_amd
+from rllm_amd.data.dataset import Dataset
+from rllm_amd.trainer.algorithms.config import AlgorithmConfig
+from rllm_amd.trainer.cpu_backend import CPUBackend
+from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+
+
+@rllm_amd.rollout
+def solve_flow(task, config):
+    r = httpx.post(config.base_url + "/chat/completions",
+                   json={"model": config.model,
+                         "messages": [{"role": "user", "content": str(task.instruction)}]},
+                   timeout=30.0)
+    r.raise_for_status()
+    return None
+
+
+@rllm_amd.evaluator
+def parity_eval(task, episode):
+    """Reward: fraction of generated token ids that are even — a dense
+    deterministic signal with within-group variance (sampling temp 1.0)."""
+    step = episode.trajectories[0].steps[-1]
+    if not step.response_ids:
+        return 0.0
+    frac = sum(1 for t in step.response_ids if t % 2 == 0) / len(step.response_ids)
+    return float(frac)
+
+
+def make_dataset(n=4):
+    return Dataset([{"question": f"task {i}", "id": str(i)} for i in range(n)]).as_tasks(id_key="id")
+
+
+def test_trainer_full_loop(tmp_path):
+    backend = CPUBackend(solve_flow, parity_eval, rollout_max_tokens=8, seed=0)
+    cfg = TrainerConfig(
+        total_epochs=1, train_batch_size=2, rollout_n=4, max_steps=2,
+        checkpoint_dir=str(tmp_path / "ckpt"), save_freq=2,
+        episode_log_dir=str(tmp_path / "episodes"), logger_backends=["jsonl"],
+    )
+    trainer = UnifiedTrainer(backend, make_dataset(4), config=cfg,
+                             algorithm_config=AlgorithmConfig())
+    trainer.fit()
+
+    assert trainer.state.global_step == 2
+    assert trainer.state.total_episodes == 2 * 2 * 4  # steps * tasks * n
+    # checkpoint written
+    assert (tmp_path / "ckpt" / "global_step_2" / "actor.pt").exists()
+    assert (tmp_path / "ckpt" / "latest_checkpointed_iteration.txt").read_text() == "2"
+    # episode logs in the canonical format
+    ep_dirs = list((tmp_path / "episodes").glob("train_step_*"))
+    assert len(ep_dirs) == 2
+    from rllm_amd.utils.episode_logger import EpisodeLogger
+
+    eps = EpisodeLogger.load_episodes(ep_dirs[0])
+    assert len(eps) == 8
+    assert all(len(e.trajectories[0].steps[0].response_ids) > 0 for e in eps)
+    # advantages were written onto steps (GRPO ran)
+    advs = [e.trajectories[0].steps[0].advantage for e in eps]
+    assert any(a not in (None, 0.0) for a in advs)
+
+
+def test_trainer_resume(tmp_path):
+    ds = make_dataset(4)
+    cfg = TrainerConfig(total_epochs=2, train_batch_size=2, rollout_n=2, max_steps=2,
+                        checkpoint_dir=str(tmp_path / "ck"), save_freq=1,
+                        logger_backends=[])
+    b1 = CPUBackend(solve_flow, parity_eval, rollout_max_tokens=6, seed=0)
+    t1 = UnifiedTrainer(b1, ds, config=cfg)
+    t1.fit()
+    assert t1.state.global_step == 2
+
+    cfg2 = TrainerConfig(total_epochs=2, train_batch_size=2, rollout_n=2, max_steps=3,
+                         checkpoint_dir=str(tmp_path / "ck"), save_freq=10,
+                         resume="auto", logger_backends=[])
+    b2 = CPUBackend(solve_flow, parity_eval, rollout_max_tokens=6, seed=0)
+    t2 = UnifiedTrainer(b2, ds, config=cfg2)
+    t2.fit()
+    assert t2.state.global_step == 3  # continued from 2
+
+
+def test_trainer_validation(tmp_path):
+    backend = CPUBackend(solve_flow, parity_eval, rollout_max_tokens=6, seed=1)
+    cfg = TrainerConfig(total_epochs=1, train_batch_size=2, rollout_n=2, max_steps=1,
+                        val_freq=1, rollout_n_val=1, logger_backends=[])
+    trainer = UnifiedTrainer(backend, make_dataset(2), val_dataset=make_dataset(2), config=cfg)
+    trainer.fit()
+    # validation ran without blowing up; metrics tracked on state
+    assert trainer.state.global_step == 1
