@@ -1,0 +1,115 @@
+"""Multi-process distributed-path tests on CPU (gloo, world_size=2):
+collective helpers + DP gradient-averaging semantics — the same code
+paths bench.py exercises over RCCL on the GPU node."""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+# spawn helpers must be module-level picklable functions
+
+
+def _setup(rank, world):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = os.environ.get("TEST_DIST_PORT", "29611")
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    from rllm_amd.parallel import dist as pdist
+
+    pdist.init_from_env(backend="gloo")
+    return pdist
+
+
+def _worker_collectives(rank, world, q):
+    try:
+        pdist = _setup(rank, world)
+        # all_reduce_sum
+        t = torch.tensor([float(rank + 1)])
+        pdist.all_reduce_sum_(t)
+        assert t.item() == 3.0, t
+        # all_reduce_max scalar
+        m = pdist.all_reduce_scalar(float(rank), op="max")
+        assert m == 1.0
+        # sum scalar
+        s = pdist.all_reduce_scalar(1.5, op="sum")
+        assert s == 3.0
+        # broadcast
+        b = torch.tensor([42.0 if rank == 0 else 0.0])
+        pdist.broadcast_(b, src=0)
+        assert b.item() == 42.0
+        # all_gather_object (C4: trajectory gather)
+        objs = pdist.all_gather_object_list({"rank": rank, "eps": [rank] * 2})
+        assert [o["rank"] for o in objs] == [0, 1]
+        pdist.barrier()
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+def _worker_dp_grad(rank, world, q):
+    """Each rank computes different grads on a shared model; after
+    all-reduce + divide both ranks hold identical averaged params."""
+    try:
+        pdist = _setup(rank, world)
+        torch.manual_seed(0)  # same init on both ranks
+        model = torch.nn.Linear(4, 2)
+        x = torch.randn(3, 4, generator=torch.Generator().manual_seed(rank + 10))
+        loss = model(x).pow(2).mean()
+        loss.backward()
+        for p in model.parameters():
+            pdist.all_reduce_sum_(p.grad)
+            p.grad /= world
+        with torch.no_grad():
+            for p in model.parameters():
+                p -= 0.1 * p.grad
+        flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+        gathered = pdist.all_gather_object_list(flat.tolist())
+        assert gathered[0] == pytest.approx(gathered[1]), "ranks diverged after DP step"
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+def _run_spawn(fn, port: str):
+    os.environ["TEST_DIST_PORT"] = port
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=fn, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(), q.get()]
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
+
+
+@pytest.mark.timeout(180)
+def test_collectives_world2():
+    _run_spawn(_worker_collectives, "29611")
+
+
+@pytest.mark.timeout(180)
+def test_dp_gradient_averaging_world2():
+    _run_spawn(_worker_dp_grad, "29613")
+
+
+def test_single_process_noop():
+    """world_size=1 path: helpers are no-ops, no process group created."""
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK"):
+        os.environ.pop(k, None)
+    from rllm_amd.parallel import dist as pdist
+
+    rank, world, local = pdist.init_from_env()
+    assert (rank, world, local) == (0, 1, 0)
+    assert pdist.all_reduce_scalar(5.0) == 5.0
+    t = torch.tensor([1.0])
+    assert pdist.all_reduce_sum_(t) is t
+    assert pdist.all_gather_object_list("x") == ["x"]
