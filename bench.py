@@ -105,11 +105,17 @@ def main():
     rng = np.random.default_rng(args.seed + 17 * rank)
     sp = SamplingParams(temperature=1.0, top_p=1.0, max_tokens=args.max_new_tokens)
 
+    phase_times = {"rollout_s": 0.0, "update_s": 0.0}
+
     def one_step(step_idx: int) -> int:
         """Returns number of response tokens generated on this rank."""
         # --- S1 rollout ---
+        t_r0 = time.monotonic()
         prompts = [rng.integers(0, cfg.vocab_size, size=args.prompt_len).tolist() for _ in range(n_seqs)]
         outs = engine.generate(prompts, sp)
+        torch.cuda.synchronize()
+        phase_times["rollout_s"] += time.monotonic() - t_r0
+        t_u0 = time.monotonic()
 
         # --- episodes with synthetic rule rewards (group-variant) ---
         episodes = []
@@ -139,11 +145,15 @@ def main():
                 return lp
         trainer.update_policy(rows, old_logprob_fn=old_lp_fn)
         engine.weight_version = trainer.weight_version  # colocated sync: shared tensors
+        torch.cuda.synchronize()
+        phase_times["update_s"] += time.monotonic() - t_u0
         return sum(len(o.token_ids) for o in outs)
 
     # --- warmup ---
     for w in range(args.warmup):
         one_step(-1 - w)
+    phase_times["rollout_s"] = 0.0
+    phase_times["update_s"] = 0.0
 
     # --- timed region ---
     pdist.barrier()
@@ -186,6 +196,8 @@ def main():
                 "temperature": 1.0, "top_p": 1.0, "clip_ratio": 0.2,
                 "kl_beta": args.kl_beta, "lr": args.lr,
                 "loss_agg": "token-mean",
+                "rollout_ms_per_step": round(phase_times["rollout_s"] / args.steps * 1000, 1),
+                "update_ms_per_step": round(phase_times["update_s"] / args.steps * 1000, 1),
                 "note": "tokens = response tokens generated AND trained on per wall-clock second, whole job",
             },
         }
