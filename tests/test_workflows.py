@@ -1,0 +1,109 @@
+"""Workflow ABC + built-ins + UnifiedWorkflowEngine against a scripted
+fake rollout engine (CPU; mirrors reference test style)."""
+
+import asyncio
+
+import pytest
+
+from rllm_amd.engine.rollout.model_output import ModelOutput
+from rllm_amd.engine.rollout.rollout_engine import RolloutEngine
+from rllm_amd.engine.unified_workflow_engine import UnifiedWorkflowEngine
+from rllm_amd.environments.frozenlake import FrozenLakeAgent, FrozenLakeEnv
+from rllm_amd.eval.runner import pass_at_k
+from rllm_amd.workflows.builtin import MultiTurnWorkflow, SimpleWorkflow
+from rllm_amd.workflows.store import InMemoryStore
+from rllm_amd.workflows.workflow import TerminationReason
+
+
+class ScriptedEngine(RolloutEngine):
+    """Returns scripted responses in order (cycling)."""
+
+    def __init__(self, responses):
+        super().__init__()
+        self.responses = responses
+        self.i = 0
+
+    async def _get_model_response(self, messages, **kwargs):
+        text = self.responses[self.i % len(self.responses)]
+        self.i += 1
+        return ModelOutput(text=text, content=text,
+                           prompt_ids=[1, 2, 3], completion_ids=[9, 9], logprobs=[-0.1, -0.1],
+                           finish_reason="stop")
+
+
+def test_simple_workflow_reward():
+    eng = ScriptedEngine(["the answer is 4"])
+
+    def reward_fn(task, response):
+        return 1.0 if "4" in response else 0.0
+
+    wf = SimpleWorkflow(eng, reward_fn=reward_fn)
+    ep = asyncio.run(wf.run_with_termination_handling({"question": "2+2?"}, "t:0"))
+    assert ep.id == "t:0"
+    assert ep.is_correct
+    assert ep.trajectories[0].reward == 1.0
+    assert ep.termination_reason == TerminationReason.ENV_DONE
+    assert ep.metrics.get("solver_acc") == 1.0
+
+
+def test_multi_turn_frozenlake_reaches_goal():
+    # path on default map: down down right right down right -> G
+    eng = ScriptedEngine(["down", "down", "right", "right", "down", "right"])
+    wf = MultiTurnWorkflow(eng, FrozenLakeAgent, FrozenLakeEnv, max_turns=10)
+    ep = asyncio.run(wf.run_with_termination_handling({}, "fl:0"))
+    assert ep.termination_reason == TerminationReason.ENV_DONE
+    traj = ep.trajectories[0]
+    assert traj.reward == 1.0
+    assert traj.steps[-1].done
+    # cumulative chat structure: later steps extend earlier ones
+    assert traj.is_cumulative()
+
+
+def test_multi_turn_hole_ends_episode():
+    eng = ScriptedEngine(["right", "down"])  # (0,1) then (1,1)=H
+    wf = MultiTurnWorkflow(eng, FrozenLakeAgent, FrozenLakeEnv, max_turns=10)
+    ep = asyncio.run(wf.run_with_termination_handling({}, "fl:1"))
+    assert ep.trajectories[0].reward == 0.0
+    assert ep.termination_reason == TerminationReason.ENV_DONE
+
+
+def test_workflow_timeout_produces_episode():
+    class SlowEngine(ScriptedEngine):
+        async def _get_model_response(self, messages, **kwargs):
+            await asyncio.sleep(5)
+            return await super()._get_model_response(messages, **kwargs)
+
+    wf = SimpleWorkflow(SlowEngine(["x"]), timeout=0.2)
+    ep = asyncio.run(wf.run_with_termination_handling({"question": "q"}, "t:0"))
+    assert ep.termination_reason == TerminationReason.TIMEOUT
+
+
+def test_unified_workflow_engine_pool():
+    eng = ScriptedEngine(["down", "down", "right", "right", "down", "right"] * 10)
+    uwe = UnifiedWorkflowEngine(
+        MultiTurnWorkflow,
+        workflow_args={"agent_cls": FrozenLakeAgent, "env_cls": FrozenLakeEnv, "max_turns": 10},
+        rollout_engine=eng, n_parallel_tasks=2)
+    eps = asyncio.run(uwe.execute_tasks([{}, {}, {}], ["a:0", "a:1", "a:2"]))
+    assert len(eps) == 3
+    assert all(e.id in ("a:0", "a:1", "a:2") for e in eps)
+
+
+def test_store():
+    async def run():
+        store = InMemoryStore()
+        await store.set("k", 1)
+        assert await store.get("k") == 1
+        await store.update("k", lambda v: (v or 0) + 1)
+        assert await store.get("k") == 2
+        await store.delete("k")
+        assert await store.get("k", "gone") == "gone"
+
+    asyncio.run(run())
+
+
+def test_pass_at_k_estimator():
+    assert pass_at_k(10, 10, 1) == 1.0
+    assert pass_at_k(10, 0, 5) == 0.0
+    assert 0 < pass_at_k(10, 3, 1) < 1
+    assert pass_at_k(10, 3, 1) == pytest.approx(0.3)
