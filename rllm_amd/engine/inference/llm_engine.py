@@ -105,7 +105,7 @@ class LLMEngine:
     def __init__(self, model: QwenModel, kv_cache: KVCache | None = None,
                  max_num_seqs: int = 1024, max_num_batched_tokens: int = 8192,
                  kv_budget_bytes: int | None = None, eos_token_id: int | None = None,
-                 seed: int = 0):
+                 seed: int = 0, use_hip_graph: bool = True, max_model_len: int = 8192):
         self.model = model
         self.cfg = model.cfg
         self.device = next(model.parameters()).device
@@ -128,7 +128,17 @@ class LLMEngine:
         self.running: list[Sequence] = []
         self.finished: dict[str, Sequence] = {}
         self._sample_step = 0
-        self._block_tables_cap = 0
+
+        # ---- hipGraph decode state (launch-bound decode: ~300 dispatches /
+        # token collapse into one graph replay; guide §1 "capture
+        # launch-bound inner loops in hipGraphs") ----
+        self.use_hip_graph = use_hip_graph
+        self.max_model_len = max_model_len
+        self._bt_width = (max_model_len + PAGE_SIZE - 1) // PAGE_SIZE
+        self._graphs: dict = {}        # (Bpad, temp) -> (graph, tokens_out, lps_out)
+        self._graph_pool = None
+        self._gb = None                # static device buffers dict
+        self._hb = None                # pinned host staging dict
 
     # ------------------------------------------------------------------
     # Request lifecycle
@@ -233,13 +243,24 @@ class LLMEngine:
         return len(input_ids)
 
     def _run_decode(self) -> int:
-        device = self.device
         batch = self.running
+        temp = batch[0].params.temperature
+        uniform = all(s.params.temperature == temp and s.params.top_k <= 0 and s.params.top_p >= 1.0
+                      for s in batch)
+        fits = all(s.total_len + 1 < self.max_model_len for s in batch)
+        if self.use_hip_graph and uniform and fits and len(batch) <= 1024:
+            n = self._run_decode_graph(batch, temp)
+        else:
+            n = self._run_decode_eager(batch)
+        self.running = [s for s in batch if s.state == SeqState.RUNNING]
+        return n
+
+    def _run_decode_eager(self, batch: list[Sequence]) -> int:
+        device = self.device
         B = len(batch)
         input_ids, positions, slot_mapping, seq_lens = [], [], [], []
         for seq in batch:
             pos = seq.total_len  # position of the NEW token
-            # grow page table if needed
             if KVCache.pages_needed(pos + 1) > len(seq.pages):
                 seq.pages.extend(self.kv.alloc(1))
             input_ids.append(seq.last_token)
@@ -259,7 +280,104 @@ class LLMEngine:
 
         hidden = self.model.forward_decode(ids_t, pos_t, self.kv, slots_t, block_tables, lens_t)
         self._sample_and_append(batch, hidden)
-        self.running = [s for s in batch if s.state == SeqState.RUNNING]
+        return B
+
+    # ------------------------------------------------------------------
+    # hipGraph decode path
+    # ------------------------------------------------------------------
+    _GRAPH_SIZES = [1, 2, 4, 8, 16, 32, 48, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
+
+    def _init_graph_buffers(self):
+        dev = self.device
+        Bmax = min(self.max_num_seqs, 1024)
+        self._gb = {
+            "tok": torch.zeros(Bmax, device=dev, dtype=torch.long),
+            "pos": torch.zeros(Bmax, device=dev, dtype=torch.int32),
+            "slot": torch.zeros(Bmax, device=dev, dtype=torch.int32),
+            "bt": torch.zeros(Bmax, self._bt_width, device=dev, dtype=torch.int32),
+            "len": torch.ones(Bmax, device=dev, dtype=torch.int32),
+            "step": torch.zeros(1, device=dev, dtype=torch.int32),
+        }
+        self._hb = {
+            "tok": torch.zeros(Bmax, dtype=torch.long, pin_memory=True),
+            "pos": torch.zeros(Bmax, dtype=torch.int32, pin_memory=True),
+            "slot": torch.zeros(Bmax, dtype=torch.int32, pin_memory=True),
+            "bt": torch.zeros(Bmax, self._bt_width, dtype=torch.int32, pin_memory=True),
+            "len": torch.ones(Bmax, dtype=torch.int32, pin_memory=True),
+        }
+
+    def _decode_compute(self, Bp: int, temp: float):
+        g = self._gb
+        hidden = self.model.forward_decode(g["tok"][:Bp], g["pos"][:Bp], self.kv,
+                                           g["slot"][:Bp], g["bt"][:Bp], g["len"][:Bp])
+        logits = self.model.logits(hidden).contiguous()
+        tokens, lps = ops.sample_logprob(logits, temp, self.seed, 0, step_tensor=g["step"])
+        return tokens, lps
+
+    def _get_graph(self, Bp: int, temp: float):
+        key = (Bp, temp)
+        if key in self._graphs:
+            return self._graphs[key]
+        # warm up eager once (cuBLAS-style workspace init), then capture
+        torch.cuda.synchronize()
+        self._decode_compute(Bp, temp)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        if self._graph_pool is None:
+            with torch.cuda.graph(graph):
+                tokens, lps = self._decode_compute(Bp, temp)
+            self._graph_pool = graph.pool()
+        else:
+            with torch.cuda.graph(graph, pool=self._graph_pool):
+                tokens, lps = self._decode_compute(Bp, temp)
+        self._graphs[key] = (graph, tokens, lps)
+        return self._graphs[key]
+
+    def _run_decode_graph(self, batch: list[Sequence], temp: float) -> int:
+        if self._gb is None:
+            self._init_graph_buffers()
+        B = len(batch)
+        Bp = next(s for s in self._GRAPH_SIZES if s >= B)
+        hb, gb = self._hb, self._gb
+
+        tok_np = hb["tok"].numpy()
+        pos_np = hb["pos"].numpy()
+        slot_np = hb["slot"].numpy()
+        len_np = hb["len"].numpy()
+        bt_np = hb["bt"].numpy()
+        for i, seq in enumerate(batch):
+            pos = seq.total_len
+            if KVCache.pages_needed(pos + 1) > len(seq.pages):
+                seq.pages.extend(self.kv.alloc(1))
+            tok_np[i] = seq.last_token
+            pos_np[i] = pos
+            slot_np[i] = seq.pages[pos // PAGE_SIZE] * PAGE_SIZE + pos % PAGE_SIZE
+            len_np[i] = pos + 1
+            np_pages = seq.pages
+            bt_np[i, : len(np_pages)] = np_pages
+        # pad rows: token 0, pos 0, slot 0 (reserved null page), len 1, bt row 0
+        if Bp > B:
+            tok_np[B:Bp] = 0
+            pos_np[B:Bp] = 0
+            slot_np[B:Bp] = 0
+            len_np[B:Bp] = 1
+            bt_np[B:Bp, 0] = 0
+
+        for k in ("tok", "pos", "slot", "len"):
+            gb[k][:Bp].copy_(hb[k][:Bp], non_blocking=True)
+        gb["bt"][:Bp].copy_(hb["bt"][:Bp], non_blocking=True)
+        gb["step"].add_(1)
+
+        graph, tokens_out, lps_out = self._get_graph(Bp, temp)
+        graph.replay()
+
+        tokens = tokens_out[:B].tolist()
+        logprobs = lps_out[:B].tolist()
+        self._sample_step += 1
+        for seq, tok, lp in zip(batch, tokens, logprobs):
+            seq.output_ids.append(int(tok))
+            seq.logprobs.append(float(lp))
+            self._maybe_finish(seq)
         return B
 
     # ------------------------------------------------------------------

@@ -259,8 +259,8 @@ def reshape_and_cache(k, v, k_pages, v_pages, slot_mapping):
     return require_ext().reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
 
 
-def sample_logprob(logits, temperature: float, seed: int, step: int):
-    return require_ext().sample_logprob(logits, temperature, seed, step)
+def sample_logprob(logits, temperature: float, seed: int, step: int, step_tensor=None):
+    return require_ext().sample_logprob(logits, temperature, seed, step, step_tensor)
 
 
 def gather_logprob(logits, tokens, temperature: float = 1.0):

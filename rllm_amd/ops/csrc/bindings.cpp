@@ -41,7 +41,8 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,
 
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
-                                          int64_t seed, int64_t step);
+                                          int64_t seed, int64_t step,
+                                          c10::optional<torch::Tensor> step_tensor);
 torch::Tensor gather_logprob(torch::Tensor logits, torch::Tensor tokens, double temperature);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

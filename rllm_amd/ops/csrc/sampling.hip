@@ -23,13 +23,15 @@ __global__ void sample_logprob_kernel(
     const uint16_t* __restrict__ logits, // [B, V] bf16
     int32_t* __restrict__ token_out,     // [B]
     float* __restrict__ logprob_out,     // [B]
+    const uint32_t* __restrict__ step_ptr, // device step counter (hipGraph-safe) or null
     int V, float inv_temp, uint32_t seed, uint32_t step, int greedy) {
   __shared__ float red_m[4], red_s[4], red_g[4];
   __shared__ int red_i[4];
 
   const int64_t row = blockIdx.x;
   const uint16_t* lr = logits + row * (int64_t)V;
-  const uint32_t row_seed = hash_u32(seed, step, (uint32_t)row);
+  const uint32_t step_eff = step_ptr ? *step_ptr : step;
+  const uint32_t row_seed = hash_u32(seed, step_eff, (uint32_t)row);
 
   // thread-local online state
   float m = -INFINITY, s = 0.f;       // logsumexp state over z
@@ -130,7 +132,8 @@ static inline hipStream_t sp_stream() {
 }
 
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
-                                          int64_t seed, int64_t step) {
+                                          int64_t seed, int64_t step,
+                                          c10::optional<torch::Tensor> step_tensor) {
   TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16 && logits.is_contiguous());
   const int64_t B = logits.size(0);
   const int V = (int)logits.size(1);
@@ -139,9 +142,14 @@ std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperatu
   auto logprobs = torch::empty({B}, logits.options().dtype(torch::kFloat32));
   const bool greedy = temperature <= 0.0;
   const float inv_temp = greedy ? 1.0f : (float)(1.0 / temperature);
+  const uint32_t* step_ptr = nullptr;
+  if (step_tensor.has_value()) {
+    TORCH_CHECK(step_tensor->dtype() == torch::kInt32 && step_tensor->is_cuda());
+    step_ptr = (const uint32_t*)step_tensor->data_ptr<int32_t>();
+  }
   hipLaunchKernelGGL(sample_logprob_kernel, dim3((unsigned)B), dim3(256), 0, sp_stream(),
                      (const uint16_t*)logits.data_ptr(), tokens.data_ptr<int32_t>(),
-                     logprobs.data_ptr<float>(), V, inv_temp,
+                     logprobs.data_ptr<float>(), step_ptr, V, inv_temp,
                      (uint32_t)seed, (uint32_t)step, greedy ? 1 : 0);
   HIP_CHECK_KERNEL();
   return {tokens, logprobs};

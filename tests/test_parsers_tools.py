@@ -1,0 +1,110 @@
+"""Chat template parsers, tool parsers, tool base, code reward."""
+
+import pytest
+
+from rllm_amd.parser.chat_template_parser import (
+    ChatTemplateParser,
+    DeepseekQwenChatTemplateParser,
+    QwenChatTemplateParser,
+)
+from rllm_amd.parser.tool_parser import QwenToolParser, R1ToolParser, ToolParser
+from rllm_amd.rewards.code_reward import RewardCodeFn, extract_code, run_tests
+from rllm_amd.tools.tool_base import MultiTool, Tool, ToolCall, ToolOutput
+from rllm_amd.utils.tokenizer import ByteTokenizer
+
+
+def test_byte_tokenizer_roundtrip():
+    tok = ByteTokenizer()
+    text = "hello <|im_start|>user\nworld<|im_end|>"
+    ids = tok.encode(text)
+    assert tok.decode(ids) == text
+    assert tok.decode(ids, skip_special_tokens=True) == "hello user\nworld"
+
+
+def test_qwen_chat_parser_format_and_masks():
+    tok = ByteTokenizer()
+    p = QwenChatTemplateParser(tok)
+    msgs = [{"role": "user", "content": "hi"}]
+    text = p.format(msgs)
+    assert text == "<|im_start|>user\nhi<|im_end|>\n<|im_start|>assistant\n"
+    ids = p.encode_messages(msgs)
+    assert tok.decode(ids) == text
+
+    msgs2 = msgs + [{"role": "assistant", "content": "yo"}]
+    ids2, mask = p.tokenize_and_mask(msgs2)
+    # assistant block tokens are masked 1
+    assert sum(mask) > 0
+    assert len(ids2) == len(mask)
+    # user tokens masked 0
+    user_len = len(tok.encode(p.format_message(msgs[0])))
+    assert all(m == 0 for m in mask[:user_len])
+
+
+def test_cumulative_tokenize_prefix_extension():
+    tok = ByteTokenizer()
+    p = QwenChatTemplateParser(tok)
+    m1 = [{"role": "user", "content": "a"}, {"role": "assistant", "content": "b"}]
+    m2 = m1 + [{"role": "user", "content": "c"}, {"role": "assistant", "content": "d"}]
+    ids, mask = p.tokenize_and_mask_cumulative([m1, m2])
+    ids_flat, mask_flat = p.tokenize_and_mask(m2)
+    assert ids == ids_flat  # prefix-extension property: same token stream
+    assert mask == mask_flat
+
+
+def test_parser_dispatch():
+    tok = ByteTokenizer()
+    assert isinstance(ChatTemplateParser.get_parser(tok, "deepseek-r1"), DeepseekQwenChatTemplateParser)
+    assert isinstance(ChatTemplateParser.get_parser(tok, "qwen2.5"), QwenChatTemplateParser)
+    assert isinstance(ToolParser.get_parser("r1"), R1ToolParser)
+    assert isinstance(ToolParser.get_parser("qwen"), QwenToolParser)
+
+
+def test_qwen_tool_parser():
+    text = 'thinking... <tool_call>\n{"name": "search", "arguments": {"q": "x"}}\n</tool_call> done'
+    calls = QwenToolParser().parse(text)
+    assert len(calls) == 1
+    assert calls[0].name == "search"
+    assert calls[0].arguments == {"q": "x"}
+
+
+def test_r1_tool_parser():
+    text = 'Sure:\n```json\n{"name": "calc", "arguments": {"expr": "1+1"}}\n```'
+    calls = R1ToolParser().parse(text)
+    assert calls[0].name == "calc"
+
+
+def test_multitool_dispatch():
+    class Adder(Tool):
+        name = "add"
+        parameters = {"type": "object", "properties": {"a": {"type": "number"}, "b": {"type": "number"}}}
+
+        def forward(self, a, b):
+            return ToolOutput(name=self.name, output=a + b)
+
+    mt = MultiTool([Adder()])
+    out = mt.execute(ToolCall(name="add", arguments={"a": 2, "b": 3}, id="c1"))
+    assert out.output == 5 and out.id == "c1"
+    out2 = mt.execute(ToolCall(name="nope", arguments={}))
+    assert out2.error
+
+
+def test_code_reward_stdin_stdout():
+    code = "x = int(input())\nprint(x * 2)"
+    tests = [{"stdin": "3", "stdout": "6"}, {"stdin": "5", "stdout": "10"}]
+    out = run_tests(code, tests)
+    assert out.is_correct and out.n_passed == 2
+
+
+def test_code_reward_assert_style_and_extraction():
+    resp = "Here's my solution:\n```python\ndef f(x):\n    return x + 1\n```"
+    assert "def f" in extract_code(resp)
+    fn = RewardCodeFn()
+    out = fn(resp, [{"assert": "assert f(2) == 3"}, {"assert": "assert f(0) == 1"}])
+    assert out.is_correct
+    out2 = fn(resp, [{"assert": "assert f(2) == 5"}])
+    assert not out2.is_correct
+
+
+def test_code_reward_timeout():
+    out = run_tests("while True: pass", [{"assert": "assert True"}], timeout=1.5)
+    assert not out.is_correct and out.error == "timeout"
