@@ -1,0 +1,77 @@
+"""SFT data prep + distillation advantage tests (CPU)."""
+
+import numpy as np
+import pytest
+
+from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+from rllm_amd.trainer.algorithms.advantage import collect_reward_and_advantage_from_trajectory_groups
+from rllm_amd.trainer.algorithms.config import AlgorithmConfig
+from rllm_amd.trainer.distill import (
+    align_cross_tokenizer,
+    compute_distill_advantages,
+    write_distill_advantages,
+)
+from rllm_amd.trainer.sft import rows_from_episodes_sft, rows_from_messages
+from rllm_amd.types import Episode, Step, Trajectory, TrajectoryGroup
+from rllm_amd.utils.tokenizer import ByteTokenizer
+
+
+def test_rows_from_messages_masks_assistant_only():
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    msgs = [{"role": "user", "content": "hi"}, {"role": "assistant", "content": "yo"}]
+    rows = rows_from_messages([msgs], parser)
+    assert len(rows) == 1
+    r = rows[0]
+    n_user = len(parser.tokenizer.encode(parser.format_message(msgs[0])))
+    assert all(m == 0 for m in r.response_mask[:n_user])
+    assert sum(r.response_mask) == len(r.tokens) - n_user
+
+
+def test_rows_from_episodes_sft_filters_incorrect():
+    def ep(correct):
+        st = Step(prompt_ids=[1, 2], response_ids=[3, 4], logprobs=[-0.1, -0.1],
+                  chat_completions=[{"role": "user", "content": "q"}])
+        return Episode(id="t:0", is_correct=correct,
+                       trajectories=[Trajectory(name="s", steps=[st], reward=1.0 if correct else 0.0)])
+
+    rows = rows_from_episodes_sft([ep(True), ep(False)])
+    assert len(rows) == 1
+
+
+def test_distill_advantage_formula():
+    s = [-1.0, -2.0, -0.5]
+    t = [-0.5, -1.0, -3.0]
+    adv = compute_distill_advantages(s, t, coef=2.0, clip=1.5)
+    np.testing.assert_allclose(adv, [1.0, 1.5, -1.5])  # 2*(0.5)=1, 2*1=2->clip 1.5, 2*(-2.5)->-1.5
+
+
+def test_distill_discounted_future_sum():
+    adv = compute_distill_advantages([0.0, 0.0], [1.0, 1.0], coef=1.0, clip=10.0, gamma=0.5)
+    # A = [1,1] -> [1 + 0.5*1, 1] = [1.5, 1]
+    np.testing.assert_allclose(adv, [1.5, 1.0])
+
+
+def test_write_distill_advantages_flows_through_precomputed_path():
+    st = Step(prompt_ids=[1], response_ids=[5, 6], logprobs=[-1.0, -1.0],
+              chat_completions=[{"role": "user", "content": "q"}])
+    ep = Episode(id="t:0", trajectories=[Trajectory(name="s", steps=[st], reward=0.0)])
+    write_distill_advantages(ep, [[-0.5, -2.0]], coef=1.0, clip=5.0)
+    assert st.advantage == pytest.approx([0.5, -1.0])
+
+    group = TrajectoryGroup(trajectories=ep.trajectories, group_id="t:s")
+    cfg = AlgorithmConfig(use_precomputed_advantage=True)
+    metrics = collect_reward_and_advantage_from_trajectory_groups([group], cfg)
+    assert st.advantage == pytest.approx([0.5, -1.0])  # untouched
+    assert "advantage/s/mean" in metrics
+
+
+def test_cross_tokenizer_alignment():
+    # student splits "hello" as "he","llo"; teacher as "hel","lo"
+    student = ["he", "llo"]
+    teacher = ["hel", "lo"]
+    t_lp = [-3.0, -2.0]
+    out = align_cross_tokenizer(student, teacher, t_lp)
+    # "he" overlaps 2/3 of "hel": -2.0 ; "llo" gets 1/3 of "hel" + all of "lo"
+    np.testing.assert_allclose(out, [-2.0, -1.0 - 2.0])
+    with pytest.raises(ValueError):
+        align_cross_tokenizer(["ab"], ["abc"], [-1.0])
