@@ -40,11 +40,11 @@ def parse_args():
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", type=str, default="r1-distill-qwen-1.5b")
-    p.add_argument("--tasks-per-gpu", type=int, default=8, help="tasks per rank per step (weak scaling)")
+    p.add_argument("--tasks-per-gpu", type=int, default=32, help="tasks per rank per step (weak scaling; 32x8=256 seqs = reference ppo_mini_batch_size)")
     p.add_argument("--rollout-n", type=int, default=8, help="samples per task (reference default 8)")
     p.add_argument("--prompt-len", type=int, default=256, help="synthetic MATH-shaped prompt length")
     p.add_argument("--max-new-tokens", type=int, default=512, help="response length cap per rollout")
-    p.add_argument("--micro-tokens", type=int, default=16384, help="ppo_max_token_len_per_gpu")
+    p.add_argument("--micro-tokens", type=int, default=32768, help="ppo_max_token_len_per_gpu")
     p.add_argument("--kl-beta", type=float, default=1e-3)
     p.add_argument("--lr", type=float, default=1e-6)
     p.add_argument("--seed", type=int, default=1234)
