@@ -113,3 +113,40 @@ def test_single_process_noop():
     t = torch.tensor([1.0])
     assert pdist.all_reduce_sum_(t) is t
     assert pdist.all_gather_object_list("x") == ["x"]
+
+
+def _worker_weight_sync(rank, world, q):
+    """Separated-mode weight broadcast: rank 0 = trainer, rank 1 = rollout."""
+    try:
+        pdist = _setup(rank, world)
+        import torch.distributed as dist
+
+        from rllm_amd.parallel.weight_sync import SeparatedWeightSync, WeightSyncGroup
+
+        flat = torch.full((64,), float(rank))  # trainer holds 0.0, rollout 1.0
+
+        class FakeEngine:
+            weight_version = 0
+            paused = False
+
+            def pause(self):
+                self.paused = True
+
+            def resume(self):
+                self.paused = False
+
+        eng = FakeEngine() if rank == 1 else None
+        ws = SeparatedWeightSync(flat, WeightSyncGroup(group=dist.group.WORLD, src_rank=0), engine=eng)
+        ws.sync(version=3)
+        assert flat.eq(0.0).all(), flat  # both ranks now hold trainer weights
+        if rank == 1:
+            assert eng.weight_version == 3 and not eng.paused
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+@pytest.mark.timeout(180)
+def test_separated_weight_sync_world2():
+    _run_spawn(_worker_weight_sync, "29615")
