@@ -1,0 +1,95 @@
+"""GPU integration: UnifiedTrainer + NativeBackend end-to-end — the full
+framework loop (gateway → flow → native engine → traces → GRPO update)
+on the MI355X substrate with a small Qwen-architecture model."""
+
+import sys
+from pathlib import Path
+
+import httpx
+import pytest
+import torch
+
+sys.path.insert(0, str(Path(__file__).parent))
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+import rllm_amd  # noqa: E402
+from rllm_amd.data.dataset import Dataset  # noqa: E402
+
+
+@rllm_amd.rollout
+def gpu_flow(task, config):
+    r = httpx.post(config.base_url + "/chat/completions",
+                   json={"model": config.model,
+                         "messages": [{"role": "user", "content": str(task.instruction)}],
+                         "max_tokens": 16},
+                   timeout=120.0)
+    r.raise_for_status()
+    return None
+
+
+@rllm_amd.evaluator
+def gpu_eval(task, episode):
+    step = episode.trajectories[0].steps[-1]
+    if not step.response_ids:
+        return 0.0
+    return float(sum(1 for t in step.response_ids if t % 2 == 0) / len(step.response_ids))
+
+
+@requires_gpu
+def test_native_backend_full_loop(tmp_path):
+    from rllm_amd.models.config import ModelConfig
+    from rllm_amd.trainer.native_backend import NativeBackend
+    from rllm_amd.trainer.policy import PolicyTrainerConfig
+    from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+
+    cfg = ModelConfig(name="gpu-loop-tiny", hidden_size=512, intermediate_size=1024,
+                      num_layers=2, num_heads=8, num_kv_heads=2, head_dim=128,
+                      vocab_size=1024, tie_word_embeddings=False)
+    backend = NativeBackend(
+        gpu_flow, gpu_eval, model_config=cfg,
+        policy_config=PolicyTrainerConfig(lr=1e-4, kl_beta=1e-3, grad_clip=1.0),
+        kv_budget_bytes=64 << 20,
+        rollout_sampling_params={"temperature": 1.0, "max_tokens": 16},
+        n_parallel_tasks=8, seed=3)
+
+    tasks = Dataset([{"question": f"task {i}", "id": str(i)} for i in range(4)]).as_tasks(id_key="id")
+    tcfg = TrainerConfig(total_epochs=1, train_batch_size=2, rollout_n=4, max_steps=2,
+                         checkpoint_dir=str(tmp_path / "ck"), save_freq=2,
+                         logger_backends=[])
+    trainer = UnifiedTrainer(backend, tasks, config=tcfg)
+    trainer.fit()
+
+    assert trainer.state.global_step == 2
+    assert trainer.state.weight_version == 2
+    # engine saw the weight-version bump (colocated sync)
+    assert backend.engine.weight_version == 2
+    # checkpoint with optimizer state exists
+    assert (tmp_path / "ck" / "global_step_2" / "actor.pt").exists()
+    # KV pages all reclaimed after training
+    assert backend.engine.kv.num_free_pages == backend.engine.kv.num_pages - 1
+
+
+@requires_gpu
+def test_sft_trainer_reduces_nll():
+    from rllm_amd.models.config import ModelConfig
+    from rllm_amd.models.qwen import QwenModel
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.trainer.sft import SFTConfig, SFTTrainer, rows_from_messages
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    cfg = ModelConfig(name="sft-tiny", hidden_size=512, intermediate_size=1024,
+                      num_layers=2, num_heads=8, num_kv_heads=2, head_dim=128,
+                      vocab_size=1024, tie_word_embeddings=False)
+    model = QwenModel(cfg, device="cuda").init_random(seed=5)
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    msgs = [[{"role": "user", "content": f"q{i}"}, {"role": "assistant", "content": "the answer is 42"}]
+            for i in range(8)]
+    rows = rows_from_messages(msgs, parser)
+    trainer = SFTTrainer(model, SFTConfig(lr=1e-3, epochs=1))
+    m1 = trainer.train_step(rows)
+    for _ in range(4):
+        m2 = trainer.train_step(rows)
+    assert m2["sft/nll"] < m1["sft/nll"], (m1, m2)
