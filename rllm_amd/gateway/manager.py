@@ -31,11 +31,12 @@ def _free_port() -> int:
 
 
 class GatewayManager:
-    def __init__(self, config: GatewayConfig | None = None, local_handler=None):
+    def __init__(self, config: GatewayConfig | None = None, local_handler=None, parser=None):
         self.config = config or GatewayConfig()
         if self.config.port == 0:
             self.config.port = _free_port()
         self.local_handler = local_handler
+        self.parser = parser
         self._server: uvicorn.Server | None = None
         self._thread: threading.Thread | None = None
         self.app = None
@@ -49,7 +50,7 @@ class GatewayManager:
 
     # ------------------------------------------------------------------
     def start(self, worker_urls: list[str] | None = None, wait: float = 15.0) -> None:
-        self.app = create_app(self.config, local_handler=self.local_handler)
+        self.app = create_app(self.config, local_handler=self.local_handler, parser=self.parser)
         uv_config = uvicorn.Config(self.app, host=self.config.host, port=self.config.port,
                                    log_level="warning", access_log=False)
         self._server = uvicorn.Server(uv_config)

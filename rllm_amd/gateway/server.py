@@ -45,7 +45,8 @@ from rllm_amd.gateway.store import make_store
 
 
 class GatewayState:
-    def __init__(self, config: GatewayConfig, local_handler: LocalHandler | None = None):
+    def __init__(self, config: GatewayConfig, local_handler: LocalHandler | None = None,
+                 parser=None):
         self.config = config
         self.store = make_store(config.store, config.sqlite_path)
         self.registry = WorkerRegistry()
@@ -53,13 +54,21 @@ class GatewayState:
         self.http = httpx.AsyncClient(timeout=config.request_timeout)
         self.proxy = ReverseProxy(self.http, max_retries=config.max_retries, local_handler=local_handler)
         self.weight_version = 0
+        self.accumulator = None
+        if config.cumulative_token_mode:
+            if parser is None:
+                raise ValueError("cumulative_token_mode requires a chat template parser")
+            from rllm_amd.gateway.token_accumulator import TokenAccumulator
+
+            self.accumulator = TokenAccumulator(parser)
 
 
 def create_app(config: GatewayConfig | None = None,
-               local_handler: LocalHandler | None = None) -> FastAPI:
+               local_handler: LocalHandler | None = None,
+               parser=None) -> FastAPI:
     config = config or GatewayConfig()
     app = FastAPI(title="rllm_amd model gateway")
-    st = GatewayState(config, local_handler=local_handler)
+    st = GatewayState(config, local_handler=local_handler, parser=parser)
     app.state.gw = st
 
     # ---- health ----
@@ -97,6 +106,8 @@ def create_app(config: GatewayConfig | None = None,
     async def delete_session(sid: str):
         n = await st.store.delete_session(sid)
         st.policy.release_session(sid)
+        if st.accumulator is not None:
+            st.accumulator.reset(sid)
         return {"deleted_traces": n}
 
     @app.post("/sessions/batch_delete")
@@ -105,6 +116,8 @@ def create_app(config: GatewayConfig | None = None,
         for sid in body.get("session_ids", []):
             total += await st.store.delete_session(sid)
             st.policy.release_session(sid)
+            if st.accumulator is not None:
+                st.accumulator.reset(sid)
         return {"deleted_traces": total}
 
     @app.get("/sessions/{sid}/traces")
@@ -159,6 +172,17 @@ def create_app(config: GatewayConfig | None = None,
         mutated = inject_params(body, session,
                                 add_logprobs=st.config.add_logprobs,
                                 add_return_token_ids=st.config.add_return_token_ids)
+
+        # cumulative token mode: rewrite turn>=2 chat requests to a
+        # pre-tokenized completions call (prefix-extension invariant)
+        cumulative = False
+        if (st.accumulator is not None and sid is not None
+                and path.endswith("chat/completions") and not mutated.get("stream")):
+            prompt_ids = st.accumulator.build_prompt_ids(sid, body.get("messages") or [])
+            if prompt_ids is not None:
+                mutated["prompt_token_ids"] = prompt_ids
+                path = "v1/completions"
+                cumulative = True
         try:
             worker = st.policy.pick(sid, st.registry.list()) if st.proxy.local_handler is None else None
         except RuntimeError as e:
@@ -203,6 +227,15 @@ def create_app(config: GatewayConfig | None = None,
             if trace.weight_version is None:
                 trace.weight_version = st.weight_version
             await st.store.add_trace(trace)
+            if st.accumulator is not None and sid is not None and (cumulative or body.get("messages")):
+                st.accumulator.record_turn(sid, body.get("messages") or [],
+                                           trace.prompt_token_ids, trace.completion_token_ids)
+            if cumulative:
+                # client sent a chat request: give back a chat-shaped body
+                for ch in resp.get("choices", []):
+                    if "message" not in ch and "text" in ch:
+                        ch["message"] = {"role": "assistant", "content": ch.pop("text")}
+                resp["object"] = "chat.completion"
             resp = strip_injected_fields(resp, client_wanted_logprobs)
         return JSONResponse(resp, status_code=status)
 

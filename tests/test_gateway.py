@@ -161,3 +161,46 @@ def test_session_sampling_param_injection_failure_recovery(stack):
     assert r.status_code == 500
     r = httpx.post(gw.base_url + "/v1/chat/completions", json={"model": "m", "messages": []})
     assert r.status_code == 200
+
+
+def test_cumulative_token_mode():
+    """Turn>=2 chat requests become pre-tokenized completions calls with the
+    prefix-extension invariant."""
+    import asyncio
+
+    from rllm_amd.gateway.native_adapter import make_torch_lm_local_handler
+    from rllm_amd.models.torch_lm import TinyTorchLM
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    model = TinyTorchLM(seed=0)
+    handler = make_torch_lm_local_handler(model, parser)
+    gw = GatewayManager(GatewayConfig(cumulative_token_mode=True),
+                        local_handler=handler, parser=parser)
+    gw.start()
+    try:
+        client = gw.client()
+        client.create_session("cum:0", sampling_params={"max_tokens": 5})
+        msgs = [{"role": "user", "content": "hi"}]
+        r1 = httpx.post(gw.session_url("cum:0") + "/chat/completions",
+                        json={"model": "m", "messages": msgs}, timeout=60.0)
+        assert r1.status_code == 200
+        reply1 = r1.json()["choices"][0]["message"]["content"]
+        msgs2 = msgs + [{"role": "assistant", "content": reply1},
+                        {"role": "user", "content": "again"}]
+        r2 = httpx.post(gw.session_url("cum:0") + "/chat/completions",
+                        json={"model": "m", "messages": msgs2}, timeout=60.0)
+        assert r2.status_code == 200
+        assert r2.json()["choices"][0]["message"]["content"] is not None
+
+        traces = client.get_traces("cum:0")
+        assert len(traces) == 2
+        t1, t2 = traces
+        # prefix-extension: turn 2 prompt == turn 1 prompt + completion + delta
+        expected_prefix = t1.prompt_token_ids + t1.completion_token_ids
+        assert t2.prompt_token_ids[: len(expected_prefix)] == expected_prefix
+        assert len(t2.prompt_token_ids) > len(expected_prefix)  # delta appended
+        client.delete_session("cum:0")
+    finally:
+        gw.stop()
