@@ -153,14 +153,25 @@ class QwenLayer(nn.Module):
         return self._finish(hidden, attn)
 
     @torch.no_grad()
-    def forward_decode(self, hidden, positions, cos_t, sin_t, kv_cache, slot_mapping,
-                       block_tables, seq_lens, layer_idx):
-        q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
+    def forward_decode_fused(self, h, delta, positions, cos_t, sin_t, kv_cache,
+                             slot_mapping, block_tables, seq_lens, layer_idx):
+        """Decode layer with fused residual chaining: returns the next delta.
+        Kernel sequence: add_rmsnorm -> qkv GEMM -> qkv_rope_cache ->
+        paged_decode -> o GEMM -> add_rmsnorm -> gate_up GEMM -> swiglu ->
+        down GEMM (9 launches/layer)."""
+        cfg = self.cfg
+        T = h.shape[0]
         k_pages, v_pages = kv_cache[layer_idx]
-        ops.reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
+        x = ops.add_rmsnorm_(h, delta, self.input_layernorm, cfg.rms_eps)
+        qkv = torch.matmul(x, self.qkv_proj.t())
+        q = ops.qkv_rope_cache(qkv, self.qkv_bias, k_pages, v_pages, cos_t, sin_t,
+                               positions, slot_mapping, cfg.num_heads, cfg.num_kv_heads)
         attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
-                                1.0 / math.sqrt(self.cfg.head_dim))
-        return self._finish(hidden, attn)
+                                1.0 / math.sqrt(cfg.head_dim))
+        attn_delta = torch.matmul(attn.reshape(T, cfg.q_size), self.o_proj.t())
+        x = ops.add_rmsnorm_(h, attn_delta, self.post_attention_layernorm, cfg.rms_eps)
+        mlp_delta = _linear(ops.swiglu(_linear(x, self.gate_up_proj)), self.down_proj)
+        return mlp_delta
 
 
 class QwenModel(nn.Module):
@@ -194,8 +205,9 @@ class QwenModel(nn.Module):
                 with torch.no_grad():
                     p.copy_(init.to(p.dtype))
             elif "bias" in name:
+                b = torch.randn(p.shape, generator=gen, dtype=torch.float32) * 0.02
                 with torch.no_grad():
-                    p.zero_()
+                    p.copy_(b.to(p.dtype))
             else:  # norms
                 with torch.no_grad():
                     p.fill_(1.0)
@@ -236,11 +248,12 @@ class QwenModel(nn.Module):
 
     @torch.no_grad()
     def forward_decode(self, input_ids, positions, kv_cache, slot_mapping, block_tables, seq_lens):
-        hidden = self.embed_tokens[input_ids]
+        h = self.embed_tokens[input_ids].contiguous()
+        delta = None
         for i, layer in enumerate(self.layers):
-            hidden = layer.forward_decode(hidden, positions, self.cos_t, self.sin_t,
-                                          kv_cache, slot_mapping, block_tables, seq_lens, i)
-        return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
+            delta = layer.forward_decode_fused(h, delta, positions, self.cos_t, self.sin_t,
+                                               kv_cache, slot_mapping, block_tables, seq_lens, i)
+        return ops.add_rmsnorm_(h, delta, self.norm, self.cfg.rms_eps)
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:

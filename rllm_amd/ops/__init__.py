@@ -259,6 +259,17 @@ def reshape_and_cache(k, v, k_pages, v_pages, slot_mapping):
     return require_ext().reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
 
 
+def qkv_rope_cache(qkv, bias, k_pages, v_pages, cos_tab, sin_tab, positions, slot_mapping, Hq, Hk):
+    """Fused bias + rope + paged KV write + contiguous-q extract (rollout path)."""
+    return require_ext().qkv_rope_cache(qkv, bias, k_pages, v_pages, cos_tab, sin_tab,
+                                        positions, slot_mapping, Hq, Hk)
+
+
+def add_rmsnorm_(h, delta, weight, eps: float = 1e-6):
+    """h += delta (in place); returns rmsnorm(h) * weight. delta may be None."""
+    return require_ext().add_rmsnorm_(h, delta, weight, eps)
+
+
 def sample_logprob(logits, temperature: float, seed: int, step: int, step_tensor=None):
     return require_ext().sample_logprob(logits, temperature, seed, step, step_tensor)
 

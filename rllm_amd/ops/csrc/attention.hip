@@ -478,8 +478,8 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor
   auto o = torch::empty_like(q);
 
   if (n_splits <= 0) {
-    // heuristic: enough blocks to fill 256 CUs twice, capped at 16
-    int target = (2 * 256) / std::max(1, B * Hk);
+    // heuristic: ~4 blocks per CU for latency hiding, capped at 16
+    int target = (4 * 256) / std::max(1, B * Hk);
     n_splits = std::min(16, std::max(1, target));
   }
   dim3 grid(B, Hk, (unsigned)n_splits);
@@ -533,4 +533,95 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,
                      (__bf16*)k_pages.data_ptr(), (__bf16*)v_pages.data_ptr(),
                      slot_mapping.data_ptr<int32_t>(), T, Hk);
   HIP_CHECK_KERNEL();
+}
+
+// ---------------------------------------------------------------------------
+// Fused QKV postprocess (decode/prefill rollout path): from the fused QKV
+// GEMM output, apply bias + rotary embedding and scatter K/V into the
+// paged cache, emitting contiguous Q — replaces split+clone+rope+
+// reshape_and_cache (4 kernels + 3 extra memory round-trips).
+// ---------------------------------------------------------------------------
+__global__ void qkv_rope_cache_kernel(
+    const __bf16* __restrict__ qkv,     // [T, (Hq+2Hk)*D]
+    const __bf16* __restrict__ bias,    // [(Hq+2Hk)*D] or null
+    __bf16* __restrict__ q_out,         // [T, Hq, D]
+    __bf16* __restrict__ Kp,            // pages [n_pages, Hk, 16, D]
+    __bf16* __restrict__ Vp,
+    const float* __restrict__ cos_tab,  // [P, D/2]
+    const float* __restrict__ sin_tab,
+    const int32_t* __restrict__ positions,    // [T]
+    const int32_t* __restrict__ slot_mapping, // [T]
+    int64_t T, int Hq, int Hk, int D) {
+  const int half = D / 2;
+  const int H_all = Hq + 2 * Hk;
+  const int64_t total = T * (int64_t)H_all * half;
+  const int row_w = H_all * D;
+
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int i = (int)(idx % half);
+    const int64_t th = idx / half;
+    const int h = (int)(th % H_all);
+    const int64_t t = th / H_all;
+
+    const int col_lo = h * D + i;
+    const int col_hi = col_lo + half;
+    float a = (float)qkv[t * row_w + col_lo];
+    float b = (float)qkv[t * row_w + col_hi];
+    if (bias) {
+      a += (float)bias[col_lo];
+      b += (float)bias[col_hi];
+    }
+
+    if (h < Hq + Hk) {  // q and k heads get rope
+      const int pos = positions[t];
+      const float c = cos_tab[(int64_t)pos * half + i];
+      const float s = sin_tab[(int64_t)pos * half + i];
+      const float ra = a * c - b * s;
+      const float rb = b * c + a * s;
+      a = ra; b = rb;
+    }
+
+    if (h < Hq) {
+      q_out[(t * Hq + h) * D + i] = (__bf16)a;
+      q_out[(t * Hq + h) * D + i + half] = (__bf16)b;
+    } else {
+      const int32_t slot = slot_mapping[t];
+      if (slot < 0) continue;
+      const int page = slot >> 4, off = slot & 15;
+      if (h < Hq + Hk) {  // k head
+        const int kh = h - Hq;
+        __bf16* dst = Kp + (((int64_t)page * Hk + kh) * PAGE_SIZE + off) * HEAD_DIM;
+        dst[i] = (__bf16)a;
+        dst[i + half] = (__bf16)b;
+      } else {            // v head (no rope)
+        const int vh = h - Hq - Hk;
+        __bf16* dst = Vp + (((int64_t)page * Hk + vh) * PAGE_SIZE + off) * HEAD_DIM;
+        dst[i] = (__bf16)a;
+        dst[i + half] = (__bf16)b;
+      }
+    }
+  }
+}
+
+torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bias,
+                             torch::Tensor k_pages, torch::Tensor v_pages,
+                             torch::Tensor cos_tab, torch::Tensor sin_tab,
+                             torch::Tensor positions, torch::Tensor slot_mapping,
+                             int64_t Hq, int64_t Hk) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
+  const int64_t T = qkv.size(0);
+  const int D = HEAD_DIM;
+  TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hk) * D);
+  auto q_out = torch::empty({T, Hq, D}, qkv.options());
+  const __bf16* bptr = bias.has_value() ? (const __bf16*)bias->data_ptr() : nullptr;
+  const int64_t total = T * (Hq + 2 * Hk) * (D / 2);
+  hipLaunchKernelGGL(qkv_rope_cache_kernel, dim3(grid_for(total, 256)), dim3(256), 0, at_stream(),
+                     (const __bf16*)qkv.data_ptr(), bptr, (__bf16*)q_out.data_ptr(),
+                     (__bf16*)k_pages.data_ptr(), (__bf16*)v_pages.data_ptr(),
+                     cos_tab.data_ptr<float>(), sin_tab.data_ptr<float>(),
+                     positions.data_ptr<int32_t>(), slot_mapping.data_ptr<int32_t>(),
+                     T, (int)Hq, (int)Hk, D);
+  HIP_CHECK_KERNEL();
+  return q_out;
 }

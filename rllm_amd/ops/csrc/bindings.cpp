@@ -2,6 +2,8 @@
 #include <torch/extension.h>
 
 // elementwise.hip
+torch::Tensor add_rmsnorm_(torch::Tensor h, c10::optional<torch::Tensor> delta,
+                           torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps, bool save_inv_rms);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w, torch::Tensor inv_rms);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_tab, torch::Tensor sin_tab,
@@ -38,6 +40,11 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor
                            int64_t n_splits);
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,
                        torch::Tensor v_pages, torch::Tensor slot_mapping);
+torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bias,
+                             torch::Tensor k_pages, torch::Tensor v_pages,
+                             torch::Tensor cos_tab, torch::Tensor sin_tab,
+                             torch::Tensor positions, torch::Tensor slot_mapping,
+                             int64_t Hq, int64_t Hk);
 
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
@@ -59,6 +66,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_prefill", &flash_prefill, "Flash causal GQA prefill (varlen, MFMA)");
   m.def("paged_decode", &paged_decode, "Paged decode attention (flash-decoding splits)");
   m.def("reshape_and_cache", &reshape_and_cache, "Scatter K/V into KV pages");
+  m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");
+  m.def("add_rmsnorm_", &add_rmsnorm_, "Fused residual add (in-place) + RMSNorm");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

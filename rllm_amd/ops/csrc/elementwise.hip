@@ -302,3 +302,82 @@ torch::Tensor swiglu_bwd(torch::Tensor dout, torch::Tensor gateup) {
   HIP_CHECK_KERNEL();
   return dgateup;
 }
+
+// ---------------------------------------------------------------------------
+// Fused residual-add + RMSNorm (decode hot path): h += delta; y = rmsnorm(h).
+// Removes one full read+write of the hidden stream per layer-norm site.
+// ---------------------------------------------------------------------------
+__global__ void add_rmsnorm_fwd_kernel(
+    uint16_t* __restrict__ h,        // [T, H] residual stream, updated in place
+    const uint16_t* __restrict__ delta, // [T, H] (may be null: plain rmsnorm)
+    const uint16_t* __restrict__ w,  // [H]
+    uint16_t* __restrict__ y,        // [T, H]
+    int H, float eps) {
+  __shared__ float scratch[16];
+  const int64_t row = blockIdx.x;
+  uint16_t* hr = h + row * (int64_t)H;
+  const uint16_t* dr = delta ? delta + row * (int64_t)H : nullptr;
+  uint16_t* yr = y + row * (int64_t)H;
+
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    short4v a = *reinterpret_cast<const short4v*>(hr + i);
+    short4v b = *reinterpret_cast<const short4v*>(hr + i + 4);
+    if (dr) {
+      short4v da = *reinterpret_cast<const short4v*>(dr + i);
+      short4v db = *reinterpret_cast<const short4v*>(dr + i + 4);
+      short4v oa, ob;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float va = bf16_to_f32((uint16_t)a[j]) + bf16_to_f32((uint16_t)da[j]);
+        float vb = bf16_to_f32((uint16_t)b[j]) + bf16_to_f32((uint16_t)db[j]);
+        oa[j] = (short)f32_to_bf16(va);
+        ob[j] = (short)f32_to_bf16(vb);
+        ss += va * va + vb * vb;
+      }
+      *reinterpret_cast<short4v*>(hr + i) = oa;
+      *reinterpret_cast<short4v*>(hr + i + 4) = ob;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float va = bf16_to_f32((uint16_t)a[j]);
+        float vb = bf16_to_f32((uint16_t)b[j]);
+        ss += va * va + vb * vb;
+      }
+    }
+  }
+  ss = block_reduce_sum(ss, scratch);
+  const float inv_rms = rsqrtf(ss / (float)H + eps);
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    short4v a = *reinterpret_cast<const short4v*>(hr + i);
+    short4v b = *reinterpret_cast<const short4v*>(hr + i + 4);
+    short4v wa = *reinterpret_cast<const short4v*>(w + i);
+    short4v wb = *reinterpret_cast<const short4v*>(w + i + 4);
+    short4v oa, ob;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      oa[j] = (short)f32_to_bf16(bf16_to_f32((uint16_t)a[j]) * inv_rms * bf16_to_f32((uint16_t)wa[j]));
+      ob[j] = (short)f32_to_bf16(bf16_to_f32((uint16_t)b[j]) * inv_rms * bf16_to_f32((uint16_t)wb[j]));
+    }
+    *reinterpret_cast<short4v*>(yr + i) = oa;
+    *reinterpret_cast<short4v*>(yr + i + 4) = ob;
+  }
+}
+
+torch::Tensor add_rmsnorm_(torch::Tensor h, c10::optional<torch::Tensor> delta,
+                           torch::Tensor w, double eps) {
+  TORCH_CHECK(h.is_cuda() && h.dtype() == torch::kBFloat16 && h.is_contiguous());
+  const int64_t T = h.numel() / h.size(-1);
+  const int H = (int)h.size(-1);
+  auto y = torch::empty_like(h);
+  const uint16_t* dptr = nullptr;
+  if (delta.has_value()) {
+    TORCH_CHECK(delta->is_contiguous() && delta->sizes() == h.sizes());
+    dptr = (const uint16_t*)delta->data_ptr();
+  }
+  hipLaunchKernelGGL(add_rmsnorm_fwd_kernel, dim3((unsigned)T), dim3(256), 0, cur_stream(),
+                     (uint16_t*)h.data_ptr(), dptr, (const uint16_t*)w.data_ptr(),
+                     (uint16_t*)y.data_ptr(), H, (float)eps);
+  HIP_CHECK_KERNEL();
+  return y;
+}
