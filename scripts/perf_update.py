@@ -33,6 +33,7 @@ def main():
     ap.add_argument("--micro-tokens", type=int, default=32768)
     ap.add_argument("--model", default="r1-distill-qwen-1.5b")
     ap.add_argument("--iters", type=int, default=2)
+    ap.add_argument("--profile-bwd", action="store_true")
     args = ap.parse_args()
 
     from rllm_amd import ops
@@ -81,6 +82,17 @@ def main():
         lp.sum().backward()
     trainer.optim.zero_grad()
     _, t_fwdbwd = timed(fwd_bwd)
+
+    if args.profile_bwd:
+        from torch.profiler import ProfilerActivity, profile
+
+        trainer.optim.zero_grad()
+        with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+            fwd_bwd()
+            torch.cuda.synchronize()
+        print(prof.key_averages().table(sort_by="self_cuda_time_total", row_limit=18,
+                                        max_name_column_width=60))
+        return
     trainer.optim.zero_grad()
     _, t_opt = timed(lambda: trainer.optim.step())
 
