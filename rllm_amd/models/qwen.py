@@ -137,9 +137,13 @@ class QwenLayer(nn.Module):
         return hidden + mlp
 
     # -- training path -------------------------------------------------------
-    def forward_train(self, hidden, positions, cu_seqlens, cos_t, sin_t):
+    def forward_train(self, hidden, positions, cu_seqlens, cos_t, sin_t, flash_tiles=None):
         q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
-        attn = train_attention(q, k, v, cu_seqlens, 1.0 / math.sqrt(self.cfg.head_dim))
+        scale = 1.0 / math.sqrt(self.cfg.head_dim)
+        if flash_tiles is not None:
+            attn = ops.flash_attention_train(q, k, v, *flash_tiles, scale)
+        else:
+            attn = train_attention(q, k, v, cu_seqlens, scale)
         return self._finish(hidden, attn)
 
     # -- rollout paths (no grad) ----------------------------------------------
@@ -214,17 +218,24 @@ class QwenModel(nn.Module):
         return self
 
     # -- training -------------------------------------------------------------
+    use_flash_training_attention: bool = True
+
     def forward_train(self, input_ids: torch.Tensor, positions: torch.Tensor,
                       cu_seqlens: list[int]) -> torch.Tensor:
         """Packed varlen forward -> final hidden states [T, H] (after norm)."""
         hidden = self.embed_tokens[input_ids]
+        flash_tiles = None
+        if self.use_flash_training_attention:
+            seqlens = [cu_seqlens[i + 1] - cu_seqlens[i] for i in range(len(cu_seqlens) - 1)]
+            flash_tiles = make_prefill_tiles(seqlens, hidden.device)
         for layer in self.layers:
             if self.gradient_checkpointing and torch.is_grad_enabled():
                 hidden = torch.utils.checkpoint.checkpoint(
                     layer.forward_train, hidden, positions, cu_seqlens,
-                    self.cos_t, self.sin_t, use_reentrant=False)
+                    self.cos_t, self.sin_t, flash_tiles, use_reentrant=False)
             else:
-                hidden = layer.forward_train(hidden, positions, cu_seqlens, self.cos_t, self.sin_t)
+                hidden = layer.forward_train(hidden, positions, cu_seqlens,
+                                             self.cos_t, self.sin_t, flash_tiles)
         return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
 
     def logprobs_for_tokens(self, input_ids, positions, cu_seqlens, targets,

@@ -251,6 +251,32 @@ def flash_prefill(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale):
     return require_ext().flash_prefill(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale)
 
 
+class _FlashAttnFn(torch.autograd.Function):
+    """Hand-written training flash attention (fwd emits LSE; bwd is the
+    FA2-style two-pass recompute — attention.hip flash_bwd_*)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale):
+        C = require_ext()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = C.flash_train_fwd(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale)
+        ctx.save_for_backward(q, k, v, o, lse, tile_seq_start, tile_row0, tile_seq_len)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        C = require_ext()
+        q, k, v, o, lse, ts, tr, tl = ctx.saved_tensors
+        dq, dk, dv = C.flash_train_bwd(q, k, v, o, dout.contiguous(), lse, ts, tr, tl, ctx.scale)
+        return dq, dk, dv, None, None, None, None
+
+
+def flash_attention_train(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale):
+    """Differentiable causal GQA flash attention (packed varlen)."""
+    return _FlashAttnFn.apply(q, k, v, tile_seq_start, tile_row0, tile_seq_len, scale)
+
+
 def paged_decode(q, k_pages, v_pages, block_tables, seq_lens, scale, n_splits: int = 0):
     return require_ext().paged_decode(q, k_pages, v_pages, block_tables, seq_lens, scale, n_splits)
 

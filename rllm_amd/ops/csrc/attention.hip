@@ -23,6 +23,15 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define HEAD_DIM 128
+
+// fwd declarations (training-attention backward kernels defined at file end)
+__global__ void attn_bwd_preprocess_kernel(const __bf16*, const __bf16*, float*, int64_t, int);
+__global__ void flash_bwd_dq_kernel(const __bf16*, const __bf16*, const __bf16*, const __bf16*,
+                                    const float*, const float*, __bf16*,
+                                    const int32_t*, const int32_t*, const int32_t*, int, int, float);
+__global__ void flash_bwd_dkv_kernel(const __bf16*, const __bf16*, const __bf16*, const __bf16*,
+                                     const float*, const float*, __bf16*, __bf16*,
+                                     const int32_t*, const int32_t*, const int32_t*, int, int, float);
 #define QTILE 64        // q rows per block (4 waves x 16)
 #define KVTILE 64       // kv tokens per inner tile
 #define PAGE_SIZE 16
@@ -51,6 +60,7 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const int32_t* __restrict__ tile_seq_start,
     const int32_t* __restrict__ tile_row0,
     const int32_t* __restrict__ tile_seq_len,
+    float* __restrict__ lse_out,              // [T, Hq] fp32 or null (training fwd)
     int Hq, int Hk, float scale) {
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
@@ -227,6 +237,9 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt) {
       orow[dt * 16 + (lane & 15)] = (__bf16)(o_acc[dt][r] * inv_l);
+    }
+    if (lse_out != nullptr && (lane & 15) == 0) {
+      lse_out[(int64_t)(seq_start + qrow) * Hq + qh] = m_st[r] + __logf(l_st[r]);
     }
   }
 }
@@ -462,9 +475,74 @@ torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                      (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(),
                      (const __bf16*)v.data_ptr(), (__bf16*)o.data_ptr(),
                      tile_seq_start.data_ptr<int32_t>(), tile_row0.data_ptr<int32_t>(),
-                     tile_seq_len.data_ptr<int32_t>(), Hq, Hk, (float)scale);
+                     tile_seq_len.data_ptr<int32_t>(), nullptr, Hq, Hk, (float)scale);
   HIP_CHECK_KERNEL();
   return o;
+}
+
+std::vector<torch::Tensor> flash_train_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                           torch::Tensor tile_seq_start, torch::Tensor tile_row0,
+                                           torch::Tensor tile_seq_len, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(q.size(2) == HEAD_DIM, "head_dim must be 128");
+  const int Hq = (int)q.size(1), Hk = (int)k.size(1);
+  const int n_tiles = (int)tile_row0.size(0);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({q.size(0), (int64_t)Hq}, q.options().dtype(torch::kFloat32));
+  dim3 grid(n_tiles, Hq);
+  hipLaunchKernelGGL(flash_prefill_kernel, grid, dim3(256), 0, at_stream(),
+                     (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(),
+                     (const __bf16*)v.data_ptr(), (__bf16*)o.data_ptr(),
+                     tile_seq_start.data_ptr<int32_t>(), tile_row0.data_ptr<int32_t>(),
+                     tile_seq_len.data_ptr<int32_t>(), lse.data_ptr<float>(),
+                     Hq, Hk, (float)scale);
+  HIP_CHECK_KERNEL();
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> flash_train_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                           torch::Tensor o, torch::Tensor dout, torch::Tensor lse,
+                                           torch::Tensor tile_seq_start, torch::Tensor tile_row0,
+                                           torch::Tensor tile_seq_len, double scale) {
+  const int64_t T = q.size(0);
+  const int Hq = (int)q.size(1), Hk = (int)k.size(1);
+  const int n_tiles = (int)tile_row0.size(0);
+  auto dout_c = dout.contiguous();
+  auto Dsum = torch::empty({T, (int64_t)Hq}, q.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(attn_bwd_preprocess_kernel, dim3((unsigned)(T * Hq)), dim3(64), 0, at_stream(),
+                     (const __bf16*)dout_c.data_ptr(), (const __bf16*)o.data_ptr(),
+                     Dsum.data_ptr<float>(), T, Hq);
+  HIP_CHECK_KERNEL();
+
+  auto dq = torch::empty_like(q);
+  auto dkh = torch::empty({T, (int64_t)Hq, (int64_t)HEAD_DIM}, q.options());
+  auto dvh = torch::empty({T, (int64_t)Hq, (int64_t)HEAD_DIM}, q.options());
+  dim3 grid(n_tiles, Hq);
+  hipLaunchKernelGGL(flash_bwd_dq_kernel, grid, dim3(256), 0, at_stream(),
+                     (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(),
+                     (const __bf16*)v.data_ptr(), (const __bf16*)dout_c.data_ptr(),
+                     lse.data_ptr<float>(), Dsum.data_ptr<float>(), (__bf16*)dq.data_ptr(),
+                     tile_seq_start.data_ptr<int32_t>(), tile_row0.data_ptr<int32_t>(),
+                     tile_seq_len.data_ptr<int32_t>(), Hq, Hk, (float)scale);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(flash_bwd_dkv_kernel, grid, dim3(256), 0, at_stream(),
+                     (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(),
+                     (const __bf16*)v.data_ptr(), (const __bf16*)dout_c.data_ptr(),
+                     lse.data_ptr<float>(), Dsum.data_ptr<float>(),
+                     (__bf16*)dkh.data_ptr(), (__bf16*)dvh.data_ptr(),
+                     tile_seq_start.data_ptr<int32_t>(), tile_row0.data_ptr<int32_t>(),
+                     tile_seq_len.data_ptr<int32_t>(), Hq, Hk, (float)scale);
+  HIP_CHECK_KERNEL();
+  // GQA: sum per-q-head partials over each group (fp32 accumulate)
+  const int G = Hq / Hk;
+  torch::Tensor dk, dv;
+  if (G > 1) {
+    dk = dkh.view({T, (int64_t)Hk, (int64_t)G, (int64_t)HEAD_DIM}).to(torch::kFloat32).sum(2).to(torch::kBFloat16);
+    dv = dvh.view({T, (int64_t)Hk, (int64_t)G, (int64_t)HEAD_DIM}).to(torch::kFloat32).sum(2).to(torch::kBFloat16);
+  } else {
+    dk = dkh; dv = dvh;
+  }
+  return {dq, dk, dv};
 }
 
 torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor v_pages,
@@ -624,4 +702,307 @@ torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bia
                      T, (int)Hq, (int)Hk, D);
   HIP_CHECK_KERNEL();
   return q_out;
+}
+
+// ===========================================================================
+// Training flash attention (fwd + bwd) — the update-path attention
+// (replaces the chunked eager softmax composition; K5/K12).
+//
+// Forward reuses the prefill structure but also emits per-row LSE
+// (m + log l, fp32) for the backward recompute. Backward is FA2-style:
+//   D[t,h]  = rowsum(dO * O)
+//   dQ pass: per q-tile, iterate kv<=q tiles: S=QK^T, P=exp(S-lse),
+//            dP=dO V^T, dS=P*(dP-D), dQ += dS K * scale
+//   dKV pass: per kv-tile, iterate q>=kv tiles: S^T=K Q^T,
+//            P^T=exp(S^T-lse_col), dV += P^T dO,
+//            dS^T = P^T*(dP^T - D_col), dK += dS^T Q * scale
+// Both passes tile exactly like the forward (MFMA 16x16x32, XOR-swizzled
+// LDS); no atomics — each output tile has one writer.
+// ===========================================================================
+
+// rowsum(dO * O) -> D [T, Hq] fp32
+__global__ void attn_bwd_preprocess_kernel(
+    const __bf16* __restrict__ dO, const __bf16* __restrict__ O,
+    float* __restrict__ D, int64_t T, int Hq) {
+  const int64_t row = blockIdx.x;  // t * Hq + h
+  if (row >= T * Hq) return;
+  const __bf16* d = dO + row * HEAD_DIM;
+  const __bf16* o = O + row * HEAD_DIM;
+  float acc = 0.f;
+  for (int i = threadIdx.x * 2; i < HEAD_DIM; i += blockDim.x * 2) {
+    acc += (float)d[i] * (float)o[i] + (float)d[i + 1] * (float)o[i + 1];
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_xor(acc, off, 64);
+  if (threadIdx.x == 0) D[row] = acc;
+}
+
+// dQ: one block per (q-tile, q-head); 4 waves x 16 q rows.
+__global__ __launch_bounds__(256, 2) void flash_bwd_dq_kernel(
+    const __bf16* __restrict__ Q, const __bf16* __restrict__ K, const __bf16* __restrict__ V,
+    const __bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dsum,
+    __bf16* __restrict__ dQ,
+    const int32_t* __restrict__ tile_seq_start, const int32_t* __restrict__ tile_row0,
+    const int32_t* __restrict__ tile_seq_len, int Hq, int Hk, float scale) {
+  const int tile = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (Hq / Hk);
+  const int seq_start = tile_seq_start[tile];
+  const int q_local0 = tile_row0[tile] - seq_start;
+  const int seq_len = tile_seq_len[tile];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wrow0 = q_local0 + wid * 16;
+
+  // LDS: K tile (linear-swz, for B of QK^T), KT (transposed, for B of dS*K),
+  //      V tile (linear-swz, for B of dO*V^T), per-wave dS staging
+  __shared__ char lds[KVTILE * HEAD_DIM * 2 + HEAD_DIM * KVTILE * 2 + KVTILE * HEAD_DIM * 2 + 4 * 16 * KVTILE * 2];
+  char* k_lds = lds;
+  char* kt_lds = lds + KVTILE * HEAD_DIM * 2;
+  char* v_lds = kt_lds + HEAD_DIM * KVTILE * 2;
+  char* ds_lds = v_lds + KVTILE * HEAD_DIM * 2;
+
+  const int a_row = wrow0 + (lane & 15);
+  const int a_k0 = (lane >> 4) * 8;
+  const bool row_valid = a_row < seq_len;
+
+  bf16x8 q_frag[4], do_frag[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+    if (row_valid) {
+      q_frag[ds] = *reinterpret_cast<const bf16x8*>(
+          Q + ((int64_t)(seq_start + a_row) * Hq + qh) * HEAD_DIM + ds * 32 + a_k0);
+      do_frag[ds] = *reinterpret_cast<const bf16x8*>(
+          dO + ((int64_t)(seq_start + a_row) * Hq + qh) * HEAD_DIM + ds * 32 + a_k0);
+    } else {
+      q_frag[ds] = bf16x8{};
+      do_frag[ds] = bf16x8{};
+    }
+  }
+  // per-lane row stats for its 4 C-layout rows
+  const int my_row0 = wrow0 + (lane >> 4) * 4;
+  float lse_r[4], d_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qr = my_row0 + r;
+    if (qr < seq_len) {
+      lse_r[r] = lse[(int64_t)(seq_start + qr) * Hq + qh];
+      d_r[r] = Dsum[(int64_t)(seq_start + qr) * Hq + qh];
+    } else { lse_r[r] = 0.f; d_r[r] = 0.f; }
+  }
+
+  f32x4 dq_acc[8];
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) dq_acc[dt] = f32x4{};
+
+  const int q_hi = min(q_local0 + QTILE, seq_len);
+  for (int kv0 = 0; kv0 < q_hi; kv0 += KVTILE) {
+    const int kv_n = min(KVTILE, seq_len - kv0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < KVTILE * (HEAD_DIM / 8); idx += 256) {
+      const int r = idx / (HEAD_DIM / 8);
+      const int c8 = (idx % (HEAD_DIM / 8)) * 8;
+      bf16x8 kv = (r < kv_n) ? *reinterpret_cast<const bf16x8*>(
+          K + ((int64_t)(seq_start + kv0 + r) * Hk + kvh) * HEAD_DIM + c8) : bf16x8{};
+      *reinterpret_cast<bf16x8*>(k_lds + swz(r, c8 * 2)) = kv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<__bf16*>(kt_lds + swz_p(c8 + j, r * 2)) = kv[j];
+      bf16x8 vv = (r < kv_n) ? *reinterpret_cast<const bf16x8*>(
+          V + ((int64_t)(seq_start + kv0 + r) * Hk + kvh) * HEAD_DIM + c8) : bf16x8{};
+      *reinterpret_cast<bf16x8*>(v_lds + swz(r, c8 * 2)) = vv;
+    }
+    __syncthreads();
+
+    // S = QK^T, dP = dO V^T per n-tile
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      f32x4 s_acc = f32x4{};
+      f32x4 dp_acc = f32x4{};
+      const int b_col = nt * 16 + (lane & 15);
+#pragma unroll
+      for (int ds = 0; ds < 4; ++ds) {
+        bf16x8 kb = *reinterpret_cast<const bf16x8*>(k_lds + swz(b_col, (ds * 32 + a_k0) * 2));
+        bf16x8 vb = *reinterpret_cast<const bf16x8*>(v_lds + swz(b_col, (ds * 32 + a_k0) * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ds], kb, s_acc, 0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[ds], vb, dp_acc, 0, 0, 0);
+      }
+      // dS = P * (dP - D); write to per-wave LDS in A-frag layout
+      char* dsw = ds_lds + wid * 16 * KVTILE * 2;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qr = my_row0 + r;
+        const int col = kv0 + nt * 16 + (lane & 15);
+        float dsv = 0.f;
+        if (qr < seq_len && col <= qr && col < seq_len) {
+          const float p = __expf(s_acc[r] * scale - lse_r[r]);
+          dsv = p * (dp_acc[r] - d_r[r]);
+        }
+        const int prow = (lane >> 4) * 4 + r;
+        *reinterpret_cast<__bf16*>(dsw + swz_p(prow, (nt * 16 + (lane & 15)) * 2)) = (__bf16)dsv;
+      }
+    }
+    __syncthreads();  // actually per-wave, but keep LDS state coherent with staging loop
+
+    // dQ += dS * K  (A = dS [16, 64kv], B = KT [kv, d])
+    char* dsw = ds_lds + wid * 16 * KVTILE * 2;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(dsw + swz_p(lane & 15, (ks * 32 + a_k0) * 2));
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) {
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(kt_lds + swz_p(dt * 16 + (lane & 15), (ks * 32 + a_k0) * 2));
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, dq_acc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // store dQ (scaled)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qr = my_row0 + r;
+    if (qr >= seq_len || qr < q_local0) continue;
+    __bf16* out = dQ + ((int64_t)(seq_start + qr) * Hq + qh) * HEAD_DIM;
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt)
+      out[dt * 16 + (lane & 15)] = (__bf16)(dq_acc[dt][r] * scale);
+  }
+}
+
+// dK/dV: one block per (kv-tile, q-head); each wave owns 16 kv rows and
+// accumulates over all q tiles >= its kv tile. Writes PER-Q-HEAD partials
+// dKh/dVh [T, Hq, D]; the wrapper sums each GQA group (no atomics).
+__global__ __launch_bounds__(256, 2) void flash_bwd_dkv_kernel(
+    const __bf16* __restrict__ Q, const __bf16* __restrict__ K, const __bf16* __restrict__ V,
+    const __bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dsum,
+    __bf16* __restrict__ dKh,   // [T, Hq, D]
+    __bf16* __restrict__ dVh,   // [T, Hq, D]
+    const int32_t* __restrict__ tile_seq_start, const int32_t* __restrict__ tile_row0,
+    const int32_t* __restrict__ tile_seq_len, int Hq, int Hk, float scale) {
+  const int tile = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (Hq / Hk);
+  const int seq_start = tile_seq_start[tile];
+  const int kv_local0 = tile_row0[tile] - seq_start;
+  const int seq_len = tile_seq_len[tile];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wrow0 = kv_local0 + wid * 16;        // this wave's kv rows
+
+  // LDS: QT 16K | Q 16K | dOT 16K | dO 16K | P^T 8K | dS^T 8K = 80 KiB
+  __shared__ char lds[4 * KVTILE * HEAD_DIM * 2 + 2 * 4 * 16 * KVTILE * 2];
+  char* qt_lds = lds;
+  char* q_lds = qt_lds + HEAD_DIM * KVTILE * 2;
+  char* dot_lds = q_lds + KVTILE * HEAD_DIM * 2;
+  char* do_lds = dot_lds + HEAD_DIM * KVTILE * 2;
+  char* pt_lds = do_lds + KVTILE * HEAD_DIM * 2;
+  char* dst_lds = pt_lds + 4 * 16 * KVTILE * 2;
+
+  const int a_row = wrow0 + (lane & 15);          // kv row for A frags
+  const int a_k0 = (lane >> 4) * 8;
+  const bool row_valid = a_row < seq_len;
+
+  bf16x8 k_frag[4], v_frag[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+    if (row_valid) {
+      k_frag[ds] = *reinterpret_cast<const bf16x8*>(
+          K + ((int64_t)(seq_start + a_row) * Hk + kvh) * HEAD_DIM + ds * 32 + a_k0);
+      v_frag[ds] = *reinterpret_cast<const bf16x8*>(
+          V + ((int64_t)(seq_start + a_row) * Hk + kvh) * HEAD_DIM + ds * 32 + a_k0);
+    } else { k_frag[ds] = bf16x8{}; v_frag[ds] = bf16x8{}; }
+  }
+
+  f32x4 dk_acc[8], dv_acc[8];
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) { dk_acc[dt] = f32x4{}; dv_acc[dt] = f32x4{}; }
+
+  const int my_kv0 = wrow0 + (lane >> 4) * 4;
+
+  // causal: only q tiles that can see this kv tile
+  const int q_start_tile = (kv_local0 / KVTILE) * KVTILE;
+  for (int q0 = q_start_tile; q0 < seq_len; q0 += KVTILE) {
+    const int q_n = min(KVTILE, seq_len - q0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < KVTILE * (HEAD_DIM / 8); idx += 256) {
+      const int r = idx / (HEAD_DIM / 8);
+      const int c8 = (idx % (HEAD_DIM / 8)) * 8;
+      bf16x8 qv = (r < q_n) ? *reinterpret_cast<const bf16x8*>(
+          Q + ((int64_t)(seq_start + q0 + r) * Hq + qh) * HEAD_DIM + c8) : bf16x8{};
+      *reinterpret_cast<bf16x8*>(q_lds + swz(r, c8 * 2)) = qv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<__bf16*>(qt_lds + swz_p(c8 + j, r * 2)) = qv[j];
+      bf16x8 dv = (r < q_n) ? *reinterpret_cast<const bf16x8*>(
+          dO + ((int64_t)(seq_start + q0 + r) * Hq + qh) * HEAD_DIM + c8) : bf16x8{};
+      *reinterpret_cast<bf16x8*>(do_lds + swz(r, c8 * 2)) = dv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<__bf16*>(dot_lds + swz_p(c8 + j, r * 2)) = dv[j];
+    }
+    __syncthreads();
+
+    // per n-tile of q cols: S^T = K Q^T, dP^T = V dO^T, then stage
+    // P^T and dS^T in per-wave LDS (A-frag layout for the next MFMAs)
+    char* ptw = pt_lds + wid * 16 * KVTILE * 2;
+    char* dstw = dst_lds + wid * 16 * KVTILE * 2;
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      f32x4 st_acc = f32x4{};
+      f32x4 dpt_acc = f32x4{};
+      const int b_col = nt * 16 + (lane & 15);  // q col within tile
+#pragma unroll
+      for (int ds = 0; ds < 4; ++ds) {
+        bf16x8 qb = *reinterpret_cast<const bf16x8*>(q_lds + swz(b_col, (ds * 32 + a_k0) * 2));
+        bf16x8 dob = *reinterpret_cast<const bf16x8*>(do_lds + swz(b_col, (ds * 32 + a_k0) * 2));
+        st_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[ds], qb, st_acc, 0, 0, 0);
+        dpt_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[ds], dob, dpt_acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kvr = my_kv0 + r;
+        const int qc = q0 + nt * 16 + (lane & 15);
+        float pt = 0.f, dst = 0.f;
+        if (kvr < seq_len && qc < seq_len && qc >= kvr) {
+          const float l = lse[(int64_t)(seq_start + qc) * Hq + qh];
+          const float dd = Dsum[(int64_t)(seq_start + qc) * Hq + qh];
+          pt = __expf(st_acc[r] * scale - l);
+          dst = pt * (dpt_acc[r] - dd);
+        }
+        const int prow = (lane >> 4) * 4 + r;
+        *reinterpret_cast<__bf16*>(ptw + swz_p(prow, (nt * 16 + (lane & 15)) * 2)) = (__bf16)pt;
+        *reinterpret_cast<__bf16*>(dstw + swz_p(prow, (nt * 16 + (lane & 15)) * 2)) = (__bf16)dst;
+      }
+    }
+
+    // dV += P^T dO ; dK += dS^T Q   (A staged per-wave; B from transposed tiles)
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(ptw + swz_p(lane & 15, (ks * 32 + a_k0) * 2));
+      bf16x8 da = *reinterpret_cast<const bf16x8*>(dstw + swz_p(lane & 15, (ks * 32 + a_k0) * 2));
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) {
+        bf16x8 dob = *reinterpret_cast<const bf16x8*>(dot_lds + swz_p(dt * 16 + (lane & 15), (ks * 32 + a_k0) * 2));
+        bf16x8 qb = *reinterpret_cast<const bf16x8*>(qt_lds + swz_p(dt * 16 + (lane & 15), (ks * 32 + a_k0) * 2));
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, dob, dv_acc[dt], 0, 0, 0);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, qb, dk_acc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // store per-q-head partials (C layout rows = kv rows)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kvr = my_kv0 + r;
+    if (kvr >= seq_len || kvr < kv_local0) continue;
+    __bf16* dko = dKh + ((int64_t)(seq_start + kvr) * Hq + qh) * HEAD_DIM;
+    __bf16* dvo = dVh + ((int64_t)(seq_start + kvr) * Hq + qh) * HEAD_DIM;
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt) {
+      dko[dt * 16 + (lane & 15)] = (__bf16)(dk_acc[dt][r] * scale);
+      dvo[dt * 16 + (lane & 15)] = (__bf16)dv_acc[dt][r];
+    }
+  }
 }

@@ -35,6 +35,13 @@ void adamw_step(torch::Tensor grad, torch::Tensor master, torch::Tensor m, torch
 torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                             torch::Tensor tile_seq_start, torch::Tensor tile_row0,
                             torch::Tensor tile_seq_len, double scale);
+std::vector<torch::Tensor> flash_train_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                           torch::Tensor tile_seq_start, torch::Tensor tile_row0,
+                                           torch::Tensor tile_seq_len, double scale);
+std::vector<torch::Tensor> flash_train_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                           torch::Tensor o, torch::Tensor dout, torch::Tensor lse,
+                                           torch::Tensor tile_seq_start, torch::Tensor tile_row0,
+                                           torch::Tensor tile_seq_len, double scale);
 torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor v_pages,
                            torch::Tensor block_tables, torch::Tensor seq_lens, double scale,
                            int64_t n_splits);
@@ -64,6 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grad_sq_sum", &grad_sq_sum, "Sum of squared gradients (bf16 flat)");
   m.def("adamw_step", &adamw_step, "Fused AdamW over flat buffers");
   m.def("flash_prefill", &flash_prefill, "Flash causal GQA prefill (varlen, MFMA)");
+  m.def("flash_train_fwd", &flash_train_fwd, "Training flash attention forward (O + LSE)");
+  m.def("flash_train_bwd", &flash_train_bwd, "Training flash attention backward (dQ,dK,dV)");
   m.def("paged_decode", &paged_decode, "Paged decode attention (flash-decoding splits)");
   m.def("reshape_and_cache", &reshape_and_cache, "Scatter K/V into KV pages");
   m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");

@@ -315,3 +315,57 @@ def test_gather_logprob():
     lp = ops.gather_logprob(logits, toks, 1.0)
     expect = torch.log_softmax(logits.float(), -1).gather(1, toks.long().unsqueeze(1)).squeeze(1)
     assert (lp - expect).abs().max() < 2e-2
+
+
+@requires_gpu
+@pytest.mark.parametrize("seqlens", [[128], [64, 200, 96]])
+def test_flash_train_fwd_bwd_vs_eager(seqlens):
+    """Hand-written training flash attention (fwd LSE + FA2-style bwd) vs
+    fp32 autograd through eager attention."""
+    torch.manual_seed(12)
+    T = sum(seqlens)
+    Hq, Hk, D = 12, 2, 128
+    scale = 1.0 / math.sqrt(D)
+    q = rand_bf16(T, Hq, D, scale=0.5)
+    k = rand_bf16(T, Hk, D, scale=0.5)
+    v = rand_bf16(T, Hk, D, scale=0.5)
+    ts, tr, tl = make_prefill_tiles(seqlens)
+
+    q1 = q.clone().requires_grad_(True)
+    k1 = k.clone().requires_grad_(True)
+    v1 = v.clone().requires_grad_(True)
+    o = ops.flash_attention_train(q1, k1, v1, ts, tr, tl, scale)
+
+    cu = [0]
+    for n in seqlens:
+        cu.append(cu[-1] + n)
+    o_ref = ref.attention_ref(q, k, v, cu, scale)
+    assert (o.float() - o_ref.float()).abs().max() < 3e-2
+
+    g = torch.randn_like(o, dtype=torch.float32).to(torch.bfloat16)
+    o.backward(g)
+
+    # fp32 reference grads
+    q2 = q.float().clone().requires_grad_(True)
+    k2 = k.float().clone().requires_grad_(True)
+    v2 = v.float().clone().requires_grad_(True)
+    G = Hq // Hk
+    loss = 0.0
+    for b in range(len(seqlens)):
+        s0, s1 = cu[b], cu[b + 1]
+        n = s1 - s0
+        ks = q2.new_zeros(0)
+        kk = k2[s0:s1].repeat_interleave(G, dim=1)
+        vv = v2[s0:s1].repeat_interleave(G, dim=1)
+        scores = torch.einsum("qhd,khd->hqk", q2[s0:s1], kk) * scale
+        mask = torch.triu(torch.ones(n, n, device=DEV, dtype=torch.bool), 1)
+        scores = scores.masked_fill(mask, float("-inf"))
+        p = torch.softmax(scores, -1)
+        oo = torch.einsum("hqk,khd->qhd", p, vv)
+        loss = loss + (oo * g[s0:s1].float()).sum()
+    loss.backward()
+
+    for got, want, name in ((q1.grad, q2.grad, "dq"), (k1.grad, k2.grad, "dk"), (v1.grad, v2.grad, "dv")):
+        err = (got.float() - want).abs().max()
+        rel = err / (want.abs().max() + 1e-6)
+        assert rel < 0.08, f"{name}: max abs err {err}, rel {rel}"
