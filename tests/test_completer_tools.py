@@ -1,0 +1,68 @@
+"""Completer helpers + python/calculator tools."""
+
+import asyncio
+
+from rllm_amd.engine.rollout.completer import Completer, TITOCompleter
+from rllm_amd.engine.rollout.model_output import ModelOutput
+from rllm_amd.engine.rollout.rollout_engine import RolloutEngine
+from rllm_amd.tools.python_tool import CalculatorTool, PythonInterpreterTool
+
+
+class EchoEngine(RolloutEngine):
+    async def _get_model_response(self, messages, **kw):
+        return ModelOutput(content=f"reply-{len(messages)}", prompt_ids=[1], completion_ids=[2], logprobs=[-0.1])
+
+
+def test_completer_keeps_history():
+    async def run():
+        c = Completer(EchoEngine(), system_prompt="sys")
+        out1 = await c("hello")
+        assert out1.content == "reply-2"
+        out2 = await c("again")
+        assert out2.content == "reply-4"
+        assert [m["role"] for m in c.messages] == ["system", "user", "assistant", "user", "assistant"]
+
+    asyncio.run(run())
+
+
+class FakeTITOEngine(RolloutEngine):
+    @property
+    def supports_token_in_token_out(self):
+        return True
+
+    async def get_token_output_from_token_input(self, token_input, **kw):
+        from rllm_amd.engine.inference.llm_engine import RequestOutput
+
+        return RequestOutput(request_id="x", prompt_ids=list(token_input),
+                             token_ids=[9, 9], logprobs=[-0.5, -0.5],
+                             finish_reason="stop", weight_version=1)
+
+    def assemble_model_output(self, token_input, token_output):
+        return ModelOutput(content="ok", prompt_ids=list(token_input),
+                           completion_ids=token_output.token_ids, logprobs=token_output.logprobs)
+
+
+def test_tito_completer_extends_stream():
+    async def run():
+        c = TITOCompleter(FakeTITOEngine())
+        out1 = await c([1, 2, 3])
+        assert out1.prompt_ids == [1, 2, 3]
+        out2 = await c([4])
+        # stream: prompt for turn 2 = turn-1 prompt + completion + new delta
+        assert out2.prompt_ids == [1, 2, 3, 9, 9, 4]
+
+    asyncio.run(run())
+
+
+def test_python_tool():
+    t = PythonInterpreterTool(timeout=10.0)
+    out = t(code="print(6*7)")
+    assert out.output.strip() == "42" and out.error is None
+    out2 = t(code="raise ValueError('x')")
+    assert out2.error and "ValueError" in out2.error
+
+
+def test_calculator_tool():
+    t = CalculatorTool()
+    assert t(expression="2*(3+4)").output == 14
+    assert t(expression="__import__('os')").error
