@@ -199,7 +199,7 @@ class LLMEngine:
         tokens = 0
         while self.waiting and len(self.running) + len(batch) < self.max_num_seqs:
             seq = self.waiting[0]
-            n = len(seq.prompt_ids)
+            n = seq.total_len
             if batch and tokens + n > self.max_num_batched_tokens:
                 break
             need = KVCache.pages_needed(n + 1)
@@ -218,8 +218,9 @@ class LLMEngine:
         input_ids, positions, slot_mapping, seqlens, last_rows = [], [], [], [], []
         row = 0
         for seq in batch:
-            n = len(seq.prompt_ids)
-            input_ids.extend(seq.prompt_ids)
+            toks = seq.prompt_ids + seq.output_ids  # resume-aware (preemption)
+            n = len(toks)
+            input_ids.extend(toks)
             positions.extend(range(n))
             slot_mapping.extend(self._slot(seq, p) for p in range(n))
             seqlens.append(n)
@@ -242,8 +243,41 @@ class LLMEngine:
                 seq.first_token_time = time.monotonic()
         return len(input_ids)
 
+    def _preempt(self, seq: Sequence) -> None:
+        """KV pressure: recycle the sequence's pages and requeue it. On
+        re-admission the prefill recomputes KV for prompt+generated-so-far
+        and decoding continues — the same mechanism gives partial-rollout
+        resume (reference async_agent_loop.py:11-14)."""
+        if seq in self.running:
+            self.running.remove(seq)
+        if seq.pages:
+            self.kv.free(seq.pages)
+            seq.pages = []
+        seq.state = SeqState.WAITING
+        self.waiting.insert(0, seq)
+        logger_warning = getattr(self, "_preempt_count", 0) + 1
+        self._preempt_count = logger_warning
+
+    def _grow_pages(self, seq: Sequence, pos: int) -> bool:
+        """Ensure capacity for position pos; False if the pool is exhausted
+        even after preempting younger sequences."""
+        while KVCache.pages_needed(pos + 1) > len(seq.pages):
+            try:
+                seq.pages.extend(self.kv.alloc(1))
+            except MemoryError:
+                victim = None
+                for cand in reversed(self.running):
+                    if cand is not seq:
+                        victim = cand
+                        break
+                if victim is None:
+                    self._preempt(seq)
+                    return False
+                self._preempt(victim)
+        return True
+
     def _run_decode(self) -> int:
-        batch = self.running
+        batch = list(self.running)  # snapshot: preemption mutates self.running
         temp = batch[0].params.temperature
         uniform = all(s.params.temperature == temp and s.params.top_k <= 0 and s.params.top_p >= 1.0
                       for s in batch)
@@ -257,17 +291,23 @@ class LLMEngine:
 
     def _run_decode_eager(self, batch: list[Sequence]) -> int:
         device = self.device
-        B = len(batch)
+        active: list[Sequence] = []
         input_ids, positions, slot_mapping, seq_lens = [], [], [], []
         for seq in batch:
+            if seq.state != SeqState.RUNNING:
+                continue  # preempted by an earlier seq's page grab
             pos = seq.total_len  # position of the NEW token
-            if KVCache.pages_needed(pos + 1) > len(seq.pages):
-                seq.pages.extend(self.kv.alloc(1))
+            if not self._grow_pages(seq, pos):
+                continue
+            active.append(seq)
             input_ids.append(seq.last_token)
             positions.append(pos)
             slot_mapping.append(self._slot(seq, pos))
             seq_lens.append(pos + 1)
-
+        if not active:
+            return 0
+        batch = active
+        B = len(batch)
         max_pages = max(len(s.pages) for s in batch)
         block_tables = torch.zeros(B, max_pages, device=device, dtype=torch.int32)
         for i, seq in enumerate(batch):
@@ -346,9 +386,11 @@ class LLMEngine:
         len_np = hb["len"].numpy()
         bt_np = hb["bt"].numpy()
         for i, seq in enumerate(batch):
+            if seq.state != SeqState.RUNNING:
+                return 0  # preempted mid-loop; retry next step with new batch
             pos = seq.total_len
-            if KVCache.pages_needed(pos + 1) > len(seq.pages):
-                seq.pages.extend(self.kv.alloc(1))
+            if not self._grow_pages(seq, pos):
+                return 0  # batch composition changed; retry next step
             tok_np[i] = seq.last_token
             pos_np[i] = pos
             slot_np[i] = seq.pages[pos // PAGE_SIZE] * PAGE_SIZE + pos % PAGE_SIZE

@@ -177,3 +177,23 @@ def test_update_policy_end_to_end():
     pos_delta = sum(after[i] - before[i] for i in range(8) if i % 2 == 1)
     neg_delta = sum(after[i] - before[i] for i in range(8) if i % 2 == 0)
     assert pos_delta > neg_delta, (pos_delta, neg_delta)
+
+
+@requires_gpu
+def test_kv_preemption_and_resume():
+    """Tiny KV pool forces preemption; every sequence still completes with
+    full-length outputs and logprobs (recompute-on-resume)."""
+    torch.manual_seed(9)
+    model = tiny_model(seed=31)
+    from rllm_amd.engine.inference.kv_cache import KVCache
+
+    # 40 pages total => ~640 token slots; 6 seqs x (32 prompt + 64 out) = 576
+    # concurrent slots needed + growth churn -> forces preemption cycles
+    kv = KVCache(model.cfg.num_layers, model.cfg.num_kv_heads, model.cfg.head_dim,
+                 40, device="cuda")
+    engine = LLMEngine(model, kv_cache=kv, eos_token_id=None, seed=4)
+    prompts = [list(range(1, 33)) for _ in range(6)]
+    outs = engine.generate(prompts, SamplingParams(temperature=1.0, max_tokens=64))
+    assert all(len(o.token_ids) == 64 for o in outs)
+    assert all(len(o.logprobs) == 64 for o in outs)
+    assert engine.kv.num_free_pages == engine.kv.num_pages - 1
