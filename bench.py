@@ -44,7 +44,8 @@ def parse_args():
     p.add_argument("--rollout-n", type=int, default=8, help="samples per task (reference default 8)")
     p.add_argument("--prompt-len", type=int, default=256, help="synthetic MATH-shaped prompt length")
     p.add_argument("--max-new-tokens", type=int, default=512, help="response length cap per rollout")
-    p.add_argument("--micro-tokens", type=int, default=32768, help="ppo_max_token_len_per_gpu")
+    p.add_argument("--micro-tokens", type=int, default=0,
+                   help="ppo_max_token_len_per_gpu (0 = auto by model size)")
     p.add_argument("--kl-beta", type=float, default=1e-3)
     p.add_argument("--lr", type=float, default=1e-6)
     p.add_argument("--seed", type=int, default=1234)
@@ -78,6 +79,12 @@ def main():
     cfg = get_model_config(args.model)
     torch.manual_seed(args.seed + rank)
 
+    if args.micro_tokens <= 0:
+        # activation memory per token scales with hidden+intermediate width;
+        # keep saved activations well under HBM alongside actor+ref+optim+KV
+        n_params_b = cfg.param_count() / 1e9
+        args.micro_tokens = 32768 if n_params_b < 3 else (8192 if n_params_b < 10 else 4096)
+
     # Actor (random init — no network for checkpoints) + frozen reference.
     # 288 GB HBM: both co-resident; rollout engine SHARES the actor weights
     # (colocated weight sync = zero-copy).
@@ -97,7 +104,7 @@ def main():
     free, _ = torch.cuda.mem_get_info()
     engine = LLMEngine(
         model, max_num_seqs=1024, max_num_batched_tokens=8192,
-        kv_budget_bytes=min(int(free * 0.5), 64 << 30),
+        kv_budget_bytes=min(int(free * 0.35), 64 << 30),
         eos_token_id=None,  # synthetic data: length-capped rollouts
         seed=args.seed * 1000 + rank)
 
