@@ -11,13 +11,20 @@ from rllm_amd.data.dataset import Dataset
 
 
 class StatefulTaskDataLoader:
+    """Seeded per-epoch shuffle + resume state. With world_size > 1 each
+    rank sees a disjoint interleaved shard of the SAME shuffled order
+    (DP data sharding; all ranks must use the same seed)."""
+
     def __init__(self, dataset: Dataset, batch_size: int, shuffle: bool = True,
-                 seed: int = 0, drop_last: bool = False):
+                 seed: int = 0, drop_last: bool = False,
+                 rank: int = 0, world_size: int = 1):
         self.dataset = dataset
         self.batch_size = batch_size
         self.shuffle = shuffle
         self.seed = seed
         self.drop_last = drop_last
+        self.rank = rank
+        self.world_size = world_size
         self.epoch = 0
         self.batch_idx = 0
 
@@ -25,6 +32,8 @@ class StatefulTaskDataLoader:
         order = list(range(len(self.dataset)))
         if self.shuffle:
             random.Random(self.seed + self.epoch).shuffle(order)
+        if self.world_size > 1:
+            order = order[self.rank :: self.world_size]
         return order
 
     def __len__(self) -> int:

@@ -88,6 +88,15 @@ class NativeBackend(BackendProtocol):
 
     # ------------------------------------------------------------------
     def init_rollout_engine(self):
+        # multi-GPU: one process per GPU (torchrun); RCCL via init_from_env
+        import os
+
+        from rllm_amd.parallel import dist as pdist
+
+        pdist.init_from_env()
+        if self.device == "cuda" and "LOCAL_RANK" in os.environ:
+            self.device = f"cuda:{os.environ['LOCAL_RANK']}"
+            torch.cuda.set_device(self.device)
         self.model = QwenModel(self.cfg, device=self.device).init_random(seed=self.seed)
         if self.checkpoint_path:
             sd = torch.load(self.checkpoint_path, weights_only=True, map_location=self.device)

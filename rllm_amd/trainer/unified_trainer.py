@@ -103,8 +103,11 @@ class UnifiedTrainer:
         self.rs_state = RejectionSamplingState()
         self.state = TrainerState()
 
+        from rllm_amd.parallel import dist as pdist
+
         self.train_loader = StatefulTaskDataLoader(
-            train_dataset, batch_size=self.config.train_batch_size, seed=self.config.seed)
+            train_dataset, batch_size=self.config.train_batch_size, seed=self.config.seed,
+            rank=pdist.get_rank(), world_size=pdist.get_world_size())
         self.val_dataset = val_dataset
         self.tracking = tracking or Tracking(
             backends=self.config.logger_backends,

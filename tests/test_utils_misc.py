@@ -87,3 +87,13 @@ def test_visualization_renders():
     out = visualize_episode(ep, color=False)
     assert "t:0" in out and "action_tokens=2" in out
     print_metrics_table({"a": 1.0, "b": 2})
+
+
+def test_dataloader_rank_sharding():
+    ds = Dataset([{"i": i} for i in range(10)])
+    dl0 = StatefulTaskDataLoader(ds, batch_size=2, seed=3, rank=0, world_size=2)
+    dl1 = StatefulTaskDataLoader(ds, batch_size=2, seed=3, rank=1, world_size=2)
+    rows0 = [r["i"] for b in dl0 for r in b]
+    rows1 = [r["i"] for b in dl1 for r in b]
+    assert sorted(rows0 + rows1) == list(range(10))
+    assert not set(rows0) & set(rows1)
