@@ -193,7 +193,7 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "DeepSeek-R1-Distill-Qwen-1.5B (random init)",
+                "model": ("DeepSeek-R1-Distill-Qwen-1.5B (random init)" if args.model == "r1-distill-qwen-1.5b" else f"{args.model} (random init)"),
                 "global_batch": n_seqs * n_gpus,
                 "seq_len": args.prompt_len + args.max_new_tokens,
                 "parallelism": f"dp{n_gpus}",
