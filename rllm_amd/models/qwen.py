@@ -167,14 +167,14 @@ class QwenLayer(nn.Module):
         T = h.shape[0]
         k_pages, v_pages = kv_cache[layer_idx]
         x = ops.add_rmsnorm_(h, delta, self.input_layernorm, cfg.rms_eps)
-        qkv = torch.matmul(x, self.qkv_proj.t())
+        qkv = ops.linear_decode(x, self.qkv_proj)
         q = ops.qkv_rope_cache(qkv, self.qkv_bias, k_pages, v_pages, cos_t, sin_t,
                                positions, slot_mapping, cfg.num_heads, cfg.num_kv_heads)
         attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
                                 1.0 / math.sqrt(cfg.head_dim))
-        attn_delta = torch.matmul(attn.reshape(T, cfg.q_size), self.o_proj.t())
+        attn_delta = ops.linear_decode(attn.reshape(T, cfg.q_size), self.o_proj)
         x = ops.add_rmsnorm_(h, attn_delta, self.post_attention_layernorm, cfg.rms_eps)
-        mlp_delta = _linear(ops.swiglu(_linear(x, self.gate_up_proj)), self.down_proj)
+        mlp_delta = ops.linear_decode(ops.swiglu(ops.linear_decode(x, self.gate_up_proj)), self.down_proj)
         return mlp_delta
 
 
@@ -268,6 +268,8 @@ class QwenModel(nn.Module):
 
     @torch.no_grad()
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        if hidden.shape[0] <= 512:
+            return ops.linear_decode(hidden, self.lm_weight)
         return torch.matmul(hidden, self.lm_weight.t())
 
     # -- HF checkpoint mapping --------------------------------------------------

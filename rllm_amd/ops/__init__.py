@@ -296,6 +296,18 @@ def add_rmsnorm_(h, delta, weight, eps: float = 1e-6):
     return require_ext().add_rmsnorm_(h, delta, weight, eps)
 
 
+def skinny_gemm(a, w, bias=None):
+    """Weight-streaming GEMM for M<=512 (decode): C = a @ w.T (+ bias)."""
+    return require_ext().skinny_gemm(a, w, bias)
+
+
+def linear_decode(x, w, bias=None):
+    """Decode-path linear: custom skinny kernel at small M, hipBLASLt above."""
+    if x.shape[0] <= 512 and x.shape[-1] % 32 == 0:
+        return require_ext().skinny_gemm(x, w, bias)
+    return torch.nn.functional.linear(x, w, bias)
+
+
 def sample_logprob(logits, temperature: float, seed: int, step: int, step_tensor=None):
     return require_ext().sample_logprob(logits, temperature, seed, step, step_tensor)
 

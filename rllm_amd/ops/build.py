@@ -25,6 +25,7 @@ SOURCES = [
     "grpo.hip",
     "adamw.hip",
     "sampling.hip",
+    "skinny_gemm.hip",
     "attention.hip",
     "bindings.cpp",
 ]

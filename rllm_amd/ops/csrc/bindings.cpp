@@ -53,6 +53,9 @@ torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bia
                              torch::Tensor positions, torch::Tensor slot_mapping,
                              int64_t Hq, int64_t Hk);
 
+// skinny_gemm.hip
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, c10::optional<torch::Tensor> bias);
+
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
                                           int64_t seed, int64_t step,
@@ -77,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reshape_and_cache", &reshape_and_cache, "Scatter K/V into KV pages");
   m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");
   m.def("add_rmsnorm_", &add_rmsnorm_, "Fused residual add (in-place) + RMSNorm");
+  m.def("skinny_gemm", &skinny_gemm, "Weight-streaming skinny-M GEMM (decode path)");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

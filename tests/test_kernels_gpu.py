@@ -369,3 +369,23 @@ def test_flash_train_fwd_bwd_vs_eager(seqlens):
         err = (got.float() - want).abs().max()
         rel = err / (want.abs().max() + 1e-6)
         assert rel < 0.08, f"{name}: max abs err {err}, rel {rel}"
+
+
+@requires_gpu
+@pytest.mark.parametrize("M,N,K,use_bias", [
+    (256, 2304, 1536, True),    # qkv
+    (256, 1536, 8960, False),   # down (K not a multiple of SK_KCHUNK)
+    (64, 1536, 1536, True),     # small N -> K-split path
+    (7, 8192, 1536, False),     # tiny batch, odd M
+    (512, 1024, 4096, True),    # two m-blocks + split
+])
+def test_skinny_gemm_vs_matmul(M, N, K, use_bias):
+    torch.manual_seed(13)
+    a = rand_bf16(M, K, scale=0.3)
+    w = rand_bf16(N, K, scale=0.3)
+    b = rand_bf16(N, scale=0.3) if use_bias else None
+    out = ops.skinny_gemm(a, w, b)
+    ref_out = torch.nn.functional.linear(a.float(), w.float(), b.float() if b is not None else None)
+    err = (out.float() - ref_out).abs().max()
+    denom = ref_out.abs().max() + 1e-6
+    assert err / denom < 0.03, f"rel err {err/denom} (abs {err})"
