@@ -48,6 +48,9 @@ def parse_args():
                    help="ppo_max_token_len_per_gpu (0 = auto by model size)")
     p.add_argument("--kl-beta", type=float, default=1e-3)
     p.add_argument("--lr", type=float, default=1e-6)
+    p.add_argument("--old-logprob-mode", default="recompute",
+                   choices=["recompute", "alias", "rollout"],
+                   help="'alias' is bit-identical with one optimizer step per batch but skips a forward")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -98,7 +101,7 @@ def main():
         PolicyTrainerConfig(
             lr=args.lr, kl_beta=args.kl_beta, eps_clip=0.2, grad_clip=1.0,
             max_tokens_per_micro=args.micro_tokens, loss_agg_mode="token-mean",
-            use_ref=args.kl_beta > 0,
+            use_ref=args.kl_beta > 0, old_logprob_mode=args.old_logprob_mode,
         ))
 
     free, _ = torch.cuda.mem_get_info()

@@ -44,6 +44,12 @@ class PolicyTrainerConfig:
     tis_mode: str | None = None     # None | "token" | "sequence"
     tis_cap: float = 5.0
     bypass_mode: bool = False        # π_old := π_rollout (skip recompute)
+    # old-logprob source: "recompute" = separate no-grad forward (reference
+    # behavior, verl_backend.py:639); "alias" = lp.detach() — bit-identical
+    # to recompute when there is exactly ONE optimizer step per mini-batch
+    # (deterministic kernels, same weights) but saves a full forward pass;
+    # "rollout" = behaves like bypass_mode.
+    old_logprob_mode: str = "recompute"
     entropy_chunk: int = 16384
     use_ref: bool = True
 
@@ -143,8 +149,10 @@ class PolicyTrainer:
                 ent_n += ent.numel()
 
             with torch.no_grad():
-                if cfg.bypass_mode or old_logprob_fn is None:
+                if cfg.bypass_mode or cfg.old_logprob_mode == "rollout" or old_logprob_fn is None:
                     old_lp = rollout_lp
+                elif cfg.old_logprob_mode == "alias":
+                    old_lp = lp.detach()
                 else:
                     old_lp = old_logprob_fn(batch, rows_idx)
                 ref_lp = None
