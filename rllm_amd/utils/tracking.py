@@ -82,6 +82,66 @@ class MlflowLogger:
         self._mlflow.end_run()
 
 
+class UILogger:
+    """Background-thread batched POST of metrics (and optionally episode
+    dumps) to a UI/collector endpoint, with heartbeats (reference
+    tracking.py:320-487). Failures never block training."""
+
+    def __init__(self, endpoint: str, run_id: str = "run", flush_interval: float = 2.0,
+                 heartbeat_interval: float = 30.0):
+        import queue
+        import threading
+
+        self.endpoint = endpoint.rstrip("/")
+        self.run_id = run_id
+        self.flush_interval = flush_interval
+        self.heartbeat_interval = heartbeat_interval
+        self._q: "queue.SimpleQueue" = queue.SimpleQueue()
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._loop, daemon=True, name="ui-logger")
+        self._thread.start()
+
+    def log(self, metrics: dict, step: int):
+        self._q.put({"type": "metrics", "run_id": self.run_id, "step": step,
+                     "metrics": {k: _plain(v) for k, v in metrics.items()}})
+
+    def log_episode(self, episode_dict: dict, step: int):
+        self._q.put({"type": "episode", "run_id": self.run_id, "step": step,
+                     "episode": episode_dict})
+
+    def _loop(self):
+        import queue
+
+        try:
+            import httpx
+
+            client = httpx.Client(timeout=5.0)
+        except Exception:  # noqa: BLE001
+            return
+        last_hb = time.time()
+        while not self._stop.is_set():
+            batch = []
+            deadline = time.time() + self.flush_interval
+            while time.time() < deadline:
+                try:
+                    batch.append(self._q.get(timeout=0.2))
+                except queue.Empty:
+                    continue
+            try:
+                if batch:
+                    client.post(self.endpoint + "/ingest", json={"events": batch})
+                if time.time() - last_hb > self.heartbeat_interval:
+                    client.post(self.endpoint + "/heartbeat", json={"run_id": self.run_id,
+                                                                    "ts": time.time()})
+                    last_hb = time.time()
+            except Exception as e:  # noqa: BLE001
+                logger.debug("UILogger post failed: %s", e)
+
+    def finish(self):
+        self._stop.set()
+        self._thread.join(timeout=5.0)
+
+
 class Tracking:
     """Fan-out logger; unknown/unavailable backends are skipped with a
     warning rather than failing the run."""
@@ -102,6 +162,8 @@ class Tracking:
                     self.loggers.append(TensorboardLogger(str(Path(log_dir) / experiment_name)))
                 elif b == "mlflow":
                     self.loggers.append(MlflowLogger(project_name, experiment_name))
+                elif b.startswith("ui:"):
+                    self.loggers.append(UILogger(b[3:], run_id=experiment_name))
                 else:
                     logger.warning("unknown tracking backend %r — skipped", b)
             except Exception as e:  # noqa: BLE001

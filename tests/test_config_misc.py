@@ -76,3 +76,53 @@ def test_curl_chat_harness_invocation_shape():
     cfg = AgentConfig(base_url="http://gw/sessions/s/v1", model="m", session_uid="s")
     cmd = h.build_invocation(task, cfg)
     assert "chat/completions" in cmd and "curl" in cmd
+
+
+def test_ui_logger_posts_batches():
+    import threading
+    import time as _time
+
+    received = []
+    from fastapi import FastAPI
+    import uvicorn
+    import socket
+
+    app = FastAPI()
+
+    @app.post("/ingest")
+    async def ingest(body: dict):
+        received.extend(body["events"])
+        return {"ok": True}
+
+    @app.post("/heartbeat")
+    async def hb(body: dict):
+        return {"ok": True}
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                           log_level="warning", access_log=False))
+    t = threading.Thread(target=server.run, daemon=True)
+    t.start()
+    import httpx
+    for _ in range(100):
+        try:
+            httpx.get(f"http://127.0.0.1:{port}/docs", timeout=1.0)
+            break
+        except Exception:
+            _time.sleep(0.05)
+
+    from rllm_amd.utils.tracking import UILogger
+
+    ui = UILogger(f"http://127.0.0.1:{port}", run_id="r1", flush_interval=0.3)
+    ui.log({"loss": 1.0}, step=0)
+    ui.log_episode({"id": "t:0"}, step=0)
+    deadline = _time.time() + 8
+    while _time.time() < deadline and len(received) < 2:
+        _time.sleep(0.1)
+    ui.finish()
+    server.should_exit = True
+    assert len(received) >= 2
+    kinds = {e["type"] for e in received}
+    assert kinds == {"metrics", "episode"}
