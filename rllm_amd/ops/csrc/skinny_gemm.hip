@@ -1,13 +1,16 @@
 // Skinny-M GEMM for the decode path: C[M,N] = A[M,K] @ W[N,K]^T (+bias).
 //
-// At decode batch sizes (M <= 512) these GEMMs are weight-stream bound:
-// the whole W matrix is read once per step while A (<1 MB) lives in L2.
-// hipBLASLt's tile choices for these shapes reach ~30% of HBM bandwidth;
-// this kernel targets the stream: each block owns a 16-column slice of W,
-// stages W tiles into XOR-swizzled LDS with coalesced loads, keeps A in
-// L2 (per-lane 16 B loads), and accumulates C[M,16] in AGPRs via MFMA
-// 16x16x32. Small-N shapes add a K-split dimension (fp32 atomics + a
-// finalize pass) so the chip stays full.
+// Decode GEMMs (M <= 512) are weight-stream bound, but hipBLASLt shows a
+// ~19-25 us fixed floor at these shapes (vs a ~1-9 us stream floor for
+// qkv/o/gate_up). Design (v2 — v1's 16-col tiles over-read A from L2 and
+// starved the MFMAs):
+//   * block = 4 waves, BN = 64 (wave w owns n-subtile w*16..w*16+15)
+//   * A chunk [256, 128] staged in LDS once per K-chunk, SHARED by all 4
+//     waves (XOR-swizzled for conflict-free ds_read_b128)
+//   * W chunk [64, 128] staged coalesced; per B-fragment each wave issues
+//     16 MFMAs (one per m-subtile) -> MFMA:ds_read ~ 1:1
+//   * small-N shapes K-split (fp32 atomics + finalize) to fill the chip
+// LDS: A 64K + W 16K = 80 KiB -> 2 blocks/CU.
 
 #include "common.hpp"
 #include <torch/extension.h>
@@ -16,101 +19,99 @@
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
-#define SK_KCHUNK 512   // K tile staged in LDS (16 rows x 512 x 2B = 16 KiB)
+#define SK_BM 256      // rows covered per block (m-block)
+#define SK_BN 64       // cols per block (16 per wave)
+#define SK_BK 128      // K chunk staged in LDS
 
-__device__ __forceinline__ int sk_swz(int row, int byte_off) {
-  return (row * (SK_KCHUNK * 2) + byte_off) ^ ((row & 7) << 4);
+// A tile [256][128] bf16, row stride 256B: XOR-swizzle for the 16-row
+// column-slice reads of the A fragments.
+__device__ __forceinline__ int a_swz(int row, int byte_off) {
+  return (row * (SK_BK * 2) + byte_off) ^ ((row & 7) << 4);
+}
+// W tile [64][128] bf16 — same geometry.
+__device__ __forceinline__ int w_swz(int row, int byte_off) {
+  return (row * (SK_BK * 2) + byte_off) ^ ((row & 7) << 4);
 }
 
-// grid: (ceil(N/16), ksplits); block: 256 threads (4 waves, each wave owns
-// 4 m-subtiles of 16 rows => block covers M<=256 rows x 16 cols).
-// For M in (256, 512]: grid.z = 2 m-blocks.
 template <bool SPLIT>
-__global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
-    const __bf16* __restrict__ A,   // [M, K]
-    const __bf16* __restrict__ W,   // [N, K]
-    const __bf16* __restrict__ bias,  // [N] or null
-    __bf16* __restrict__ C,         // [M, N] (direct path)
-    float* __restrict__ C32,        // [M, N] fp32 (split path, pre-zeroed)
+__global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
+    const __bf16* __restrict__ A,    // [M, K]
+    const __bf16* __restrict__ W,    // [N, K]
+    const __bf16* __restrict__ bias, // [N] or null
+    __bf16* __restrict__ C,          // [M, N] (direct)
+    float* __restrict__ C32,         // [M, N] (split, pre-zeroed)
     int M, int N, int K) {
-  const int n0 = blockIdx.x * 16;
+  const int n0 = blockIdx.x * SK_BN;
   const int ksplit = blockIdx.y;
   const int nsplits = gridDim.y;
-  const int m_blk = blockIdx.z * 256;
-  if (n0 >= N) return;
+  const int m_blk = blockIdx.z * SK_BM;
 
-  const int k_per = ((K + nsplits - 1) / nsplits + SK_KCHUNK - 1) / SK_KCHUNK * SK_KCHUNK;
+  const int k_per = ((K + nsplits - 1) / nsplits + SK_BK - 1) / SK_BK * SK_BK;
   const int k_lo = ksplit * k_per;
   const int k_hi = min(K, k_lo + k_per);
 
-  __shared__ char w_lds[2][16 * SK_KCHUNK * 2];
+  __shared__ char a_lds[SK_BM * SK_BK * 2];  // 64 KiB
+  __shared__ char w_lds[SK_BN * SK_BK * 2];  // 16 KiB
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int a_k0 = (lane >> 4) * 8;     // k offset within a 32-slice
-  const int col = lane & 15;            // n within tile / m row selector
+  const int a_k0 = (lane >> 4) * 8;
+  const int col = lane & 15;
 
-  f32x4_t acc[4];
+  f32x4_t acc[16];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) acc[i] = f32x4_t{};
+  for (int i = 0; i < 16; ++i) acc[i] = f32x4_t{};
 
-  int buf = 0;
-  // prologue: stage first W chunk (coalesced along K)
-  {
-    const int kc0 = k_lo;
-    for (int idx = threadIdx.x; idx < 16 * (SK_KCHUNK / 8); idx += 256) {
-      const int r = idx / (SK_KCHUNK / 8);
-      const int c8 = (idx % (SK_KCHUNK / 8)) * 8;
+  for (int kc = k_lo; kc < k_hi; kc += SK_BK) {
+    __syncthreads();
+    // stage A [256, 128]: 256x128/8 = 4096 vec8 slots, 16 per thread
+    for (int idx = threadIdx.x; idx < SK_BM * (SK_BK / 8); idx += 256) {
+      const int r = idx / (SK_BK / 8);
+      const int c8 = (idx % (SK_BK / 8)) * 8;
+      bf16x8_t av{};
+      if (m_blk + r < M && kc + c8 < K) {
+        av = *reinterpret_cast<const bf16x8_t*>(A + (int64_t)(m_blk + r) * K + kc + c8);
+      }
+      *reinterpret_cast<bf16x8_t*>(&a_lds[0] + a_swz(r, c8 * 2)) = av;
+    }
+    // stage W [64, 128]: 1024 vec8 slots, 4 per thread
+    for (int idx = threadIdx.x; idx < SK_BN * (SK_BK / 8); idx += 256) {
+      const int r = idx / (SK_BK / 8);
+      const int c8 = (idx % (SK_BK / 8)) * 8;
       bf16x8_t wv{};
-      if (n0 + r < N && kc0 + c8 < k_hi) {
-        wv = *reinterpret_cast<const bf16x8_t*>(W + (int64_t)(n0 + r) * K + kc0 + c8);
+      if (n0 + r < N && kc + c8 < K) {
+        wv = *reinterpret_cast<const bf16x8_t*>(W + (int64_t)(n0 + r) * K + kc + c8);
       }
-      *reinterpret_cast<bf16x8_t*>(&w_lds[0][0] + sk_swz(r, c8 * 2)) = wv;
+      *reinterpret_cast<bf16x8_t*>(&w_lds[0] + w_swz(r, c8 * 2)) = wv;
     }
-  }
-  __syncthreads();
+    __syncthreads();
 
-  for (int kc = k_lo; kc < k_hi; kc += SK_KCHUNK, buf ^= 1) {
-    // prefetch next chunk into the other buffer
-    const int kn = kc + SK_KCHUNK;
-    if (kn < k_hi) {
-      for (int idx = threadIdx.x; idx < 16 * (SK_KCHUNK / 8); idx += 256) {
-        const int r = idx / (SK_KCHUNK / 8);
-        const int c8 = (idx % (SK_KCHUNK / 8)) * 8;
-        bf16x8_t wv{};
-        if (n0 + r < N && kn + c8 < k_hi) {
-          wv = *reinterpret_cast<const bf16x8_t*>(W + (int64_t)(n0 + r) * K + kn + c8);
-        }
-        *reinterpret_cast<bf16x8_t*>(&w_lds[buf ^ 1][0] + sk_swz(r, c8 * 2)) = wv;
-      }
-    }
-
-    const int kend = min(SK_KCHUNK, k_hi - kc);
-    for (int ks = 0; ks < kend; ks += 32) {
-      bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
-          &w_lds[buf][0] + sk_swz(col, (ks + a_k0) * 2));
+    const int kend = min(SK_BK, k_hi - kc);
 #pragma unroll
-      for (int mt = 0; mt < 4; ++mt) {
-        const int m = m_blk + wid * 64 + mt * 16 + col;
-        bf16x8_t a_frag{};
-        if (m < M && kc + ks + a_k0 < K) {
-          a_frag = *reinterpret_cast<const bf16x8_t*>(A + (int64_t)m * K + kc + ks + a_k0);
-        }
+    for (int ks = 0; ks < SK_BK; ks += 32) {
+      if (ks >= kend) break;
+      // this wave's B fragment: W rows wid*16 + col, k slice ks..ks+32
+      bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
+          &w_lds[0] + w_swz(wid * 16 + col, (ks + a_k0) * 2));
+#pragma unroll
+      for (int mt = 0; mt < 16; ++mt) {
+        bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(
+            &a_lds[0] + a_swz(mt * 16 + col, (ks + a_k0) * 2));
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc[mt], 0, 0, 0);
       }
     }
-    __syncthreads();
   }
 
-  // epilogue: C layout row=(lane>>4)*4+r within each m-subtile, col=lane&15
-  const float bias_v = (bias && n0 + col < N) ? (float)bias[n0 + col] : 0.f;
+  // epilogue: C row = mt*16 + (lane>>4)*4 + r, col = n0 + wid*16 + (lane&15)
+  const int n = n0 + wid * 16 + col;
+  if (n >= N) return;
+  const float bias_v = bias ? (float)bias[n] : 0.f;
 #pragma unroll
-  for (int mt = 0; mt < 4; ++mt) {
+  for (int mt = 0; mt < 16; ++mt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int m = m_blk + wid * 64 + mt * 16 + (lane >> 4) * 4 + r;
-      const int n = n0 + col;
-      if (m >= M || n >= N) continue;
+      const int m = m_blk + mt * 16 + (lane >> 4) * 4 + r;
+      if (m >= M) continue;
       if (SPLIT) {
         atomicAdd(&C32[(int64_t)m * N + n], acc[mt][r]);
       } else {
@@ -146,13 +147,11 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, c10::optional<torch:
   auto c = torch::empty({M, N}, a.options());
   const __bf16* bptr = bias.has_value() ? (const __bf16*)bias->data_ptr() : nullptr;
 
-  const int n_tiles = (N + 15) / 16;
-  const int m_blocks = (M + 255) / 256;
-  // enough blocks to fill the chip; round K-splits to chunk granularity
+  const int n_tiles = (N + SK_BN - 1) / SK_BN;
+  const int m_blocks = (M + SK_BM - 1) / SK_BM;
   int ksplits = 1;
-  const int kchunks = (K + SK_KCHUNK - 1) / SK_KCHUNK;
-  while (ksplits < kchunks && n_tiles * m_blocks * ksplits < 512) ksplits *= 2;
-  ksplits = std::min(ksplits, kchunks);
+  const int kchunks = (K + SK_BK - 1) / SK_BK;
+  while (ksplits * 2 <= kchunks && n_tiles * m_blocks * ksplits < 512) ksplits *= 2;
 
   dim3 grid(n_tiles, ksplits, m_blocks);
   if (ksplits == 1) {
