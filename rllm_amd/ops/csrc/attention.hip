@@ -32,6 +32,9 @@ __global__ void flash_bwd_dq_kernel(const __bf16*, const __bf16*, const __bf16*,
 __global__ void flash_bwd_dkv_kernel(const __bf16*, const __bf16*, const __bf16*, const __bf16*,
                                      const float*, const float*, __bf16*, __bf16*,
                                      const int32_t*, const int32_t*, const int32_t*, int, int, float);
+__global__ void paged_decode_mfma_kernel(const __bf16*, const __bf16*, const __bf16*, __bf16*,
+                                         float*, float*, const int32_t*, const int32_t*,
+                                         int, int, int, int, float);
 #define QTILE 64        // q rows per block (4 waves x 16)
 #define KVTILE 64       // kv tokens per inner tile
 #define PAGE_SIZE 16
@@ -571,6 +574,21 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor
     ws_ml_ptr = ws_ml.data_ptr<float>();
   }
 
+  if (G <= 16) {
+    hipLaunchKernelGGL(paged_decode_mfma_kernel, grid, dim3(256), 0, at_stream(),
+                       (const __bf16*)q.data_ptr(), (const __bf16*)k_pages.data_ptr(),
+                       (const __bf16*)v_pages.data_ptr(), (__bf16*)o.data_ptr(),
+                       ws_o_ptr, ws_ml_ptr,
+                       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+                       Hq, Hk, G, max_pages, (float)scale);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      hipLaunchKernelGGL(decode_merge_kernel, dim3(B * Hq), dim3(64), 0, at_stream(),
+                         ws_o_ptr, ws_ml_ptr, (__bf16*)o.data_ptr(), Hq, (int)n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return o;
+  }
 #define LAUNCH_G(GV) \
   hipLaunchKernelGGL(paged_decode_kernel<GV>, grid, dim3(256), 0, at_stream(), \
                      (const __bf16*)q.data_ptr(), (const __bf16*)k_pages.data_ptr(), \
@@ -1003,6 +1021,234 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_kernel(
     for (int dt = 0; dt < 8; ++dt) {
       dko[dt * 16 + (lane & 15)] = (__bf16)(dk_acc[dt][r] * scale);
       dvo[dt * 16 + (lane & 15)] = (__bf16)dv_acc[dt][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Paged decode v2 (MFMA): the VALU version spends ~200 VALU ops per KV token
+// on GQA dot products; here the GQA group (G<=16 q heads, padded to the
+// MFMA M-tile) attends to page PAIRS (2x16 tokens = one 16-col MFMA B tile
+// each) — QK^T 8 MFMAs + PV 8 MFMAs per 32 tokens, K read direct from
+// pages (one 16 B load per lane), V staged transposed per wave.
+// Grid (B, Hk, splits); block = 4 waves, each wave owns a token quarter;
+// two-level merge (wave LDS + split workspace) as in v1.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void paged_decode_mfma_kernel(
+    const __bf16* __restrict__ Q,
+    const __bf16* __restrict__ Kp,
+    const __bf16* __restrict__ Vp,
+    __bf16* __restrict__ O,
+    float* __restrict__ ws_o,
+    float* __restrict__ ws_ml,
+    const int32_t* __restrict__ block_tables,
+    const int32_t* __restrict__ seq_lens,
+    int Hq, int Hk, int G, int max_pages, float scale) {
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int n_splits = gridDim.z;
+  const int seq_len = seq_lens[b];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+
+  // token range for this (split, wave): page-pair (32-token) granularity
+  const int pairs_total = (seq_len + 31) / 32;
+  const int pairs_per_split = (pairs_total + n_splits - 1) / n_splits;
+  const int pair_lo_split = split * pairs_per_split;
+  const int pair_hi_split = min(pairs_total, pair_lo_split + pairs_per_split);
+  const int pairs_per_wave = (pair_hi_split - pair_lo_split + 3) / 4;
+  const int pair_lo = pair_lo_split + wid * pairs_per_wave;
+  const int pair_hi = min(pair_hi_split, pair_lo + pairs_per_wave);
+
+  const int32_t* bt = block_tables + (int64_t)b * max_pages;
+
+  // per-wave LDS: VT [128][32] bf16 (8 KiB) + P [16][32] bf16 (1 KiB)
+  __shared__ char vt_lds_all[4][128 * 32 * 2];
+  __shared__ char p_lds_all[4][16 * 32 * 2];
+  __shared__ float merge_o[4][16][HEAD_DIM];   // per-wave o partials (16 KiB... fp32 = 32 KiB)
+  __shared__ float merge_ml[4][16][2];
+  char* vt_lds = vt_lds_all[wid];
+  char* p_lds = p_lds_all[wid];
+
+  const int a_k0 = (lane >> 4) * 8;
+  const int col = lane & 15;
+
+  // Q fragments: A[16 rows = q heads (padded), 32 dims per slice]
+  bf16x8 q_frag[4];
+  const int qh = kvh * G + col;  // row 'col' of the M tile
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+    if (col < G) {
+      q_frag[ds] = *reinterpret_cast<const bf16x8*>(
+          Q + ((int64_t)b * Hq + qh) * HEAD_DIM + ds * 32 + a_k0);
+    } else {
+      q_frag[ds] = bf16x8{};
+    }
+  }
+
+  float m_st[4], l_st[4];
+  f32x4 o_acc[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_st[r] = -INFINITY; l_st[r] = 0.f; }
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) o_acc[dt] = f32x4{};
+
+  for (int pp = pair_lo; pp < pair_hi; ++pp) {
+    const int t0 = pp * 32;
+    // ---- QK^T over the two pages ----
+    f32x4 s_acc[2];
+#pragma unroll
+    for (int pg = 0; pg < 2; ++pg) {
+      s_acc[pg] = f32x4{};
+      const int tok = t0 + pg * 16 + col;       // this lane's B column
+      const int tok_c = min(tok, seq_len - 1);
+      const int page = bt[tok_c >> 4];
+      const __bf16* krow = Kp + (((int64_t)page * Hk + kvh) * PAGE_SIZE + (tok_c & 15)) * HEAD_DIM;
+#pragma unroll
+      for (int ds = 0; ds < 4; ++ds) {
+        bf16x8 kb = *reinterpret_cast<const bf16x8*>(krow + ds * 32 + a_k0);
+        s_acc[pg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ds], kb, s_acc[pg], 0, 0, 0);
+      }
+    }
+    // ---- stage V^T for the pair (each lane: 2 tokens x 8 dims -> 16 scalar writes) ----
+    {
+      const int tok_base = t0 + (lane >> 4) * 4;  // 4 waves of 16 lanes... use 64 lanes = 32 tok x 2 halves
+      // simpler mapping: lane covers token (lane & 31), dim half (lane >> 5)
+      const int tok = t0 + (lane & 31);
+      const int half = (lane >> 5) * 64;          // dims 0..63 or 64..127
+      const int tok_c = min(tok, seq_len - 1);
+      const int page = bt[tok_c >> 4];
+      const __bf16* vrow = Vp + (((int64_t)page * Hk + kvh) * PAGE_SIZE + (tok_c & 15)) * HEAD_DIM + half;
+#pragma unroll
+      for (int c8 = 0; c8 < 64; c8 += 8) {
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(vrow + c8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          // VT[dim][tok], row stride 32 tokens * 2B = 64B, swizzled
+          const int d = half + c8 + j;
+          const int byte = (d * 32 + (tok - t0)) * 2;
+          *reinterpret_cast<__bf16*>(vt_lds + (byte ^ ((d & 7) << 4))) = vv[j];
+        }
+      }
+    }
+    // ---- softmax (online) ----
+    const int my_r0 = (lane >> 4) * 4;
+    float p_val[2][4];
+    float row_max[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_max[r] = -INFINITY;
+#pragma unroll
+    for (int pg = 0; pg < 2; ++pg) {
+      const int tok = t0 + pg * 16 + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = s_acc[pg][r] * scale;
+        if (tok >= seq_len) s = -INFINITY;
+        p_val[pg][r] = s;
+        row_max[r] = fmaxf(row_max[r], s);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        row_max[r] = fmaxf(row_max[r], __shfl_xor(row_max[r], off, 64));
+      const float m_new = fmaxf(m_st[r], row_max[r]);
+      const float rs = (m_st[r] == -INFINITY) ? 0.f : __expf(m_st[r] - m_new);
+      float row_sum = 0.f;
+#pragma unroll
+      for (int pg = 0; pg < 2; ++pg) {
+        float e = (p_val[pg][r] == -INFINITY) ? 0.f : __expf(p_val[pg][r] - m_new);
+        p_val[pg][r] = e;
+        row_sum += e;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) row_sum += __shfl_xor(row_sum, off, 64);
+      l_st[r] = l_st[r] * rs + row_sum;
+      m_st[r] = m_new;
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) o_acc[dt][r] *= rs;
+    }
+    // ---- write P to LDS in A-frag layout [q=16][tok=32] ----
+#pragma unroll
+    for (int pg = 0; pg < 2; ++pg) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int prow = my_r0 + r;
+        const int pcol = pg * 16 + col;
+        const int byte = (prow * 32 + pcol) * 2;
+        *reinterpret_cast<__bf16*>(p_lds + (byte ^ ((prow & 7) << 4))) = (__bf16)p_val[pg][r];
+      }
+    }
+    // ---- PV: A = P [16, 32], B = VT (dim tiles) ----
+    bf16x8 a_frag;
+    {
+      const int prow = col;
+      const int byte = (prow * 32 + a_k0) * 2;
+      a_frag = *reinterpret_cast<const bf16x8*>(p_lds + (byte ^ ((prow & 7) << 4)));
+    }
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt) {
+      const int d = dt * 16 + col;
+      const int byte = (d * 32 + a_k0) * 2;
+      bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(vt_lds + (byte ^ ((d & 7) << 4)));
+      o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, o_acc[dt], 0, 0, 0);
+    }
+  }
+
+  // ---- merge 4 waves via LDS (rows = q heads 0..15 in C layout) ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = (lane >> 4) * 4 + r;
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt) {
+      merge_o[wid][qrow][dt * 16 + col] = o_acc[dt][r];
+    }
+    if (col == 0) {
+      merge_ml[wid][qrow][0] = m_st[r];
+      merge_ml[wid][qrow][1] = l_st[r];
+    }
+  }
+  __syncthreads();
+  if (wid == 0) {
+    // 64 lanes: lane covers (q row = lane>>2 & 15? ) — use 16 rows x 4 dim-blocks
+    const int qrow = lane >> 2;          // 0..15
+    const int dblk = (lane & 3) * 32;    // 4 x 32 dims
+    if (qrow < 16) {
+      float m_tot = -INFINITY;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) m_tot = fmaxf(m_tot, merge_ml[w][qrow][0]);
+      float l_tot = 0.f;
+      float o_tot[32];
+#pragma unroll
+      for (int j = 0; j < 32; ++j) o_tot[j] = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        const float mw = merge_ml[w][qrow][0];
+        const float rr = (mw == -INFINITY) ? 0.f : __expf(mw - m_tot);
+        l_tot += merge_ml[w][qrow][1] * rr;
+#pragma unroll
+        for (int j = 0; j < 32; ++j) o_tot[j] += merge_o[w][qrow][dblk + j] * rr;
+      }
+      if (qrow < G) {
+        const int qh_out = kvh * G + qrow;
+        if (n_splits == 1) {
+          const float inv_l = (l_tot > 0.f) ? 1.f / l_tot : 0.f;
+          __bf16* orow = O + ((int64_t)b * Hq + qh_out) * HEAD_DIM + dblk;
+#pragma unroll
+          for (int j = 0; j < 32; ++j) orow[j] = (__bf16)(o_tot[j] * inv_l);
+        } else {
+          float* wrow = ws_o + (((int64_t)b * Hq + qh_out) * n_splits + split) * HEAD_DIM + dblk;
+#pragma unroll
+          for (int j = 0; j < 32; ++j) wrow[j] = o_tot[j];
+          if (dblk == 0) {
+            float* mlrow = ws_ml + (((int64_t)b * Hq + qh_out) * n_splits + split) * 2;
+            mlrow[0] = m_tot;
+            mlrow[1] = l_tot;
+          }
+        }
+      }
     }
   }
 }
