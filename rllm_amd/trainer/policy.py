@@ -127,6 +127,7 @@ class PolicyTrainer:
             return {"actor/skipped": 1.0}
 
         self.optim.zero_grad()
+        offpolicy_metrics: dict = {}
         tot_loss = 0.0
         tot_pg = 0.0
         tot_kl_count = 0
@@ -162,6 +163,10 @@ class PolicyTrainer:
                                                     tgt, chunk=cfg.entropy_chunk, want_entropy=False)
                 tis_w = self._tis_weights(old_lp, rollout_lp)
 
+            if micro is micros[0]:
+                from rllm_amd.utils.offpolicy import compute_offpolicy_metrics
+
+                offpolicy_metrics = compute_offpolicy_metrics(old_lp, rollout_lp)
             eps_hi = cfg.eps_clip_high if cfg.eps_clip_high is not None else cfg.eps_clip
             loss_tok, clipped = ops.grpo_loss_per_token(
                 lp, old_lp, ref_lp, adv, tis_w,
@@ -204,6 +209,7 @@ class PolicyTrainer:
             "actor/n_micro_batches": len(micros),
             "actor/n_response_tokens": n_local_tokens,
         }
+        metrics.update(offpolicy_metrics)
         return metrics
 
     # ------------------------------------------------------------------

@@ -97,3 +97,26 @@ def test_dataloader_rank_sharding():
     rows1 = [r["i"] for b in dl1 for r in b]
     assert sorted(rows0 + rows1) == list(range(10))
     assert not set(rows0) & set(rows1)
+
+
+def test_offpolicy_metrics():
+    old = torch.tensor([-1.0, -2.0, -0.5])
+    roll = torch.tensor([-1.1, -1.9, -0.6])
+    from rllm_amd.utils.offpolicy import compute_offpolicy_metrics
+
+    m = compute_offpolicy_metrics(old, roll)
+    assert m["offpolicy/kl"] >= 0
+    assert 0.9 < m["offpolicy/ppl_ratio"] < 1.1
+    assert m["offpolicy/pearson"] > 0.9
+    # identical -> zero KL, pearson 1
+    m2 = compute_offpolicy_metrics(old, old.clone())
+    assert m2["offpolicy/kl"] == pytest.approx(0.0, abs=1e-6)
+
+
+def test_split_thought():
+    from rllm_amd.system_prompts import split_thought
+
+    th, ans = split_thought("<think>hmm 2+2</think>4")
+    assert th == "hmm 2+2" and ans == "4"
+    th2, ans2 = split_thought("just 4")
+    assert th2 == "" and ans2 == "just 4"
