@@ -1,6 +1,7 @@
 """Dataset builders, timing helpers, visualization, dataloader resume."""
 
 import pytest
+import torch
 
 from rllm_amd.data.builders import synthetic_countdown, synthetic_gsm8k, synthetic_mcq
 from rllm_amd.data.dataloader import StatefulTaskDataLoader, interleave_tasks
