@@ -79,3 +79,35 @@ def test_sandbox_task_hooks_provisions_env():
         assert mgr.active == {}
 
     asyncio.run(run())
+
+
+def test_snapshot_registry_dedup_refcount_ttl():
+    async def run():
+        from rllm_amd.sandbox.snapshot import SnapshotRegistry, snapshot_key
+
+        created, destroyed = [], []
+
+        async def create(spec):
+            created.append(spec)
+            return f"img-{len(created)}"
+
+        async def destroy(handle):
+            destroyed.append(handle)
+
+        reg = SnapshotRegistry(create, destroy, ttl_s=0.0)  # instant expiry when unref'd
+        spec = {"image": "py:3.11", "setup": ["pip install x"]}
+        k1, h1 = await reg.acquire(spec)
+        k2, h2 = await reg.acquire(spec)
+        assert k1 == k2 and h1 == h2 and len(created) == 1  # dedup
+        k3, _ = await reg.acquire({"image": "other"})
+        assert k3 != k1 and len(created) == 2
+
+        await reg.release(k1)
+        assert reg.stats()["snapshots"] >= 1  # still referenced once
+        await reg.release(k1)
+        # both refs gone + ttl 0 -> evicted and destroyed
+        await reg.release(k3)
+        assert "img-1" in destroyed
+        assert snapshot_key(spec) == snapshot_key(dict(reversed(list(spec.items())))), "order-stable hash"
+
+    asyncio.run(run())
