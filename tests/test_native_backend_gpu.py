@@ -93,3 +93,92 @@ def test_sft_trainer_reduces_nll():
     for _ in range(4):
         m2 = trainer.train_step(rows)
     assert m2["sft/nll"] < m1["sft/nll"], (m1, m2)
+
+
+@rllm_amd.rollout
+def two_turn_flow(task, config):
+    """Multi-turn agent: two chat calls with accumulated history (the
+    FrozenLake/tool-agent shape, BASELINE config 4)."""
+    msgs = [{"role": "user", "content": f"move for {task.instruction}?"}]
+    r1 = httpx.post(config.base_url + "/chat/completions",
+                    json={"model": config.model, "messages": msgs, "max_tokens": 8},
+                    timeout=120.0)
+    r1.raise_for_status()
+    reply = r1.json()["choices"][0]["message"]["content"]
+    msgs = msgs + [{"role": "assistant", "content": reply},
+                   {"role": "user", "content": "grid changed. next move?"}]
+    r2 = httpx.post(config.base_url + "/chat/completions",
+                    json={"model": config.model, "messages": msgs, "max_tokens": 8},
+                    timeout=120.0)
+    r2.raise_for_status()
+    return None
+
+
+@requires_gpu
+def test_multiturn_cumulative_rloo_native(tmp_path):
+    """Config 4 shape: multi-turn flow -> gateway (cumulative token mode) ->
+    native engine; RLOO advantages; prefix-merged packed rows in the update."""
+    from rllm_amd.gateway.models import GatewayConfig
+    from rllm_amd.models.config import ModelConfig
+    from rllm_amd.trainer.algorithms.config import AlgorithmConfig
+    from rllm_amd.trainer.native_backend import NativeBackend
+    from rllm_amd.trainer.policy import PolicyTrainerConfig
+    from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+
+    cfg = ModelConfig(name="mt-tiny", hidden_size=512, intermediate_size=1024,
+                      num_layers=2, num_heads=8, num_kv_heads=2, head_dim=128,
+                      vocab_size=1024, tie_word_embeddings=False)
+    backend = NativeBackend(
+        two_turn_flow, gpu_eval, model_config=cfg,
+        policy_config=PolicyTrainerConfig(lr=1e-4, kl_beta=0.0),
+        kv_budget_bytes=64 << 20,
+        rollout_sampling_params={"temperature": 1.0, "max_tokens": 8},
+        n_parallel_tasks=4, seed=11)
+    # cumulative token mode on the backend's gateway
+    backend.init_rollout_engine()
+    backend.gateway.stop()
+    from rllm_amd.gateway.manager import GatewayManager
+    from rllm_amd.gateway.native_adapter import make_native_local_handler
+
+    handler = make_native_local_handler(backend.driver, backend.parser, cfg.name)
+    backend.gateway = GatewayManager(GatewayConfig(cumulative_token_mode=True),
+                                     local_handler=handler, parser=backend.parser)
+    backend.gateway.start()
+    from rllm_amd.engine.agentflow_engine import AgentFlowEngine
+
+    backend.flow_engine = AgentFlowEngine(
+        two_turn_flow, backend.gateway, model_name=cfg.name, evaluator=gpu_eval,
+        n_parallel_tasks=4, default_sampling_params={"temperature": 1.0, "max_tokens": 8})
+
+    from rllm_amd.data.dataset import Dataset
+
+    tasks = Dataset([{"question": f"t{i}", "id": str(i)} for i in range(2)]).as_tasks(id_key="id")
+    tcfg = TrainerConfig(total_epochs=1, train_batch_size=2, rollout_n=4, max_steps=1,
+                         logger_backends=[], episode_log_dir=str(tmp_path / "eps"))
+    trainer = UnifiedTrainer(backend, tasks, config=tcfg,
+                             algorithm_config=AlgorithmConfig(estimator="rloo"))
+
+    # intercept the backend batch to verify prefix merging
+    seen_rows = []
+    orig = backend.transform_to_backend_batch
+
+    def capture(groups):
+        rows = orig(groups)
+        seen_rows.extend(rows)
+        return rows
+
+    backend.transform_to_backend_batch = capture
+    trainer.fit()
+
+    assert trainer.state.global_step == 1
+    assert seen_rows, "no training rows produced"
+    # every episode had 2 turns; cumulative mode must merge them into ONE
+    # packed row with two action segments (prefix-extension held)
+    merged = [r for r in seen_rows if sum(r.response_mask) >= 16]
+    assert merged, f"no merged two-turn rows (rows={[(len(r.tokens), sum(r.response_mask)) for r in seen_rows]})"
+    row = merged[0]
+    # mask pattern: 0s (prompt) 1s (turn1) 0s (obs) 1s (turn2)
+    import itertools
+
+    segs = [k for k, _ in itertools.groupby(row.response_mask)]
+    assert segs == [0, 1, 0, 1], segs
