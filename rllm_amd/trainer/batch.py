@@ -120,7 +120,10 @@ def _step_row(step: Step) -> PackedRow | None:
 
 
 def rows_from_trajectory(traj: Trajectory) -> list[PackedRow]:
-    if len(traj.steps) > 1 and traj.is_cumulative():
+    # The token chain is the ground truth for cumulativeness: attempt the
+    # prefix merge whenever there are multiple steps; _merge_cumulative
+    # returns None (-> per-step fallback) if the ids don't chain.
+    if len(traj.steps) > 1:
         merged = _merge_cumulative(traj)
         if merged is not None:
             return [merged]

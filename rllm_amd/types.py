@@ -346,11 +346,16 @@ class Trajectory:
     def is_cumulative(self) -> bool:
         """True when each step's chat_completions extends the previous step's
         as an exact prefix (reference types.py:301-314)."""
+        def key(msgs):
+            # compare on (role, content): traces may carry extra fields
+            # (reasoning, tool_calls) that don't affect cumulativeness
+            return [(m.get("role"), m.get("content")) for m in msgs]
+
         prev = None
         for step in self.steps:
             if prev is not None:
-                prev_cc = prev.chat_completions
-                curr_cc = step.chat_completions
+                prev_cc = key(prev.chat_completions)
+                curr_cc = key(step.chat_completions)
                 if not (len(curr_cc) >= len(prev_cc) and curr_cc[: len(prev_cc)] == prev_cc):
                     return False
             prev = step
