@@ -55,6 +55,8 @@ class CPUBackend(BackendProtocol):
 
     # ------------------------------------------------------------------
     def init_rollout_engine(self):
+        if self.flow_engine is not None:  # idempotent
+            return self.flow_engine
         handler = make_torch_lm_local_handler(
             self.model, self.parser, eos_token_id=None, seed=self.seed,
             weight_version_ref=self.weight_version_ref)

@@ -88,6 +88,8 @@ class NativeBackend(BackendProtocol):
 
     # ------------------------------------------------------------------
     def init_rollout_engine(self):
+        if self.rollout_engine is not None:  # idempotent (trainer calls this too)
+            return self.rollout_engine
         # multi-GPU: one process per GPU (torchrun); RCCL via init_from_env
         import os
 
