@@ -105,3 +105,77 @@ class MultiTurnWorkflow(Workflow):
             termination = TerminationReason.ENV_DONE
         self.commit(agent=self.agent)
         return self.postprocess_episode(self.collect_trajectories(), termination)
+
+
+class CumulativeWorkflow(Workflow):
+    """Multi-turn conversation where every turn extends the same message
+    history (reference cumulative_workflow.py): pairs with the gateway's
+    cumulative token mode so turn N's prompt ids prefix-extend turn N-1.
+    env_fn(task, history) -> (observation, reward, done)."""
+
+    def __init__(self, rollout_engine, env_fn, max_turns: int = 8,
+                 system_prompt: str | None = None, **kwargs):
+        super().__init__(rollout_engine, **kwargs)
+        self.env_fn = env_fn
+        self.max_turns = max_turns
+        self.system_prompt = system_prompt
+
+    async def run(self, task: dict, uid: str, **kwargs) -> Episode | None:
+        self.reset(task, uid)
+        messages: list[dict] = []
+        if self.system_prompt:
+            messages.append({"role": "system", "content": self.system_prompt})
+        obs, reward, done = self.env_fn(task, [])
+        steps: list[Step] = []
+        termination = TerminationReason.MAX_TURNS_EXCEEDED
+        for _turn in range(self.max_turns):
+            messages.append({"role": "user", "content": str(obs)})
+            out = await self.rollout_engine.get_model_response(messages, **kwargs)
+            messages.append({"role": "assistant", "content": out.content or ""})
+            step = Step.from_model_output(out, messages=messages[:-1])
+            obs, reward, done = self.env_fn(task, messages)
+            step.reward = float(reward)
+            step.done = bool(done)
+            steps.append(step)
+            if done:
+                termination = TerminationReason.ENV_DONE
+                break
+        traj = Trajectory(name="solver", steps=steps)
+        self.commit(trajectory=traj)
+        return self.postprocess_episode(self.collect_trajectories(), termination)
+
+
+class DistillationWorkflow(Workflow):
+    """On-policy distillation rollout (reference distillation_workflow.py):
+    the student samples; the teacher engine scores the student's tokens;
+    per-token advantages = coef*(teacher_lp - student_lp) flow through the
+    precomputed-advantage path."""
+
+    def __init__(self, rollout_engine, teacher_engine, coef: float = 1.0,
+                 clip: float = 5.0, **kwargs):
+        super().__init__(rollout_engine, **kwargs)
+        self.teacher_engine = teacher_engine
+        self.coef = coef
+        self.clip = clip
+
+    async def run(self, task: dict, uid: str, **kwargs) -> Episode | None:
+        from rllm_amd.trainer.distill import compute_distill_advantages
+
+        self.reset(task, uid)
+        instruction = task.get("question") or task.get("prompt") or str(task)
+        messages = [{"role": "user", "content": instruction}]
+        out = await self.rollout_engine.get_model_response(messages, **kwargs)
+        step = Step.from_model_output(out, messages=messages)
+        step.done = True
+        # teacher scores the student's sampled tokens (TITO when available)
+        teacher_lp = None
+        if getattr(self.teacher_engine, "supports_token_in_token_out", False) and out.prompt_ids:
+            scored = await self.teacher_engine.get_token_output_from_token_input(
+                list(out.prompt_ids), scored_token_ids=list(out.completion_ids or []), **kwargs)
+            teacher_lp = getattr(scored, "logprobs", None)
+        if teacher_lp is not None and step.logprobs:
+            step.advantage = compute_distill_advantages(
+                step.logprobs, teacher_lp, coef=self.coef, clip=self.clip)
+        traj = Trajectory(name="student", steps=[step])
+        self.commit(trajectory=traj)
+        return self.postprocess_episode(self.collect_trajectories(), TerminationReason.ENV_DONE)

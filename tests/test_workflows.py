@@ -107,3 +107,44 @@ def test_pass_at_k_estimator():
     assert pass_at_k(10, 0, 5) == 0.0
     assert 0 < pass_at_k(10, 3, 1) < 1
     assert pass_at_k(10, 3, 1) == pytest.approx(0.3)
+
+
+def test_cumulative_workflow_builds_chain():
+    from rllm_amd.workflows.builtin import CumulativeWorkflow
+
+    eng = ScriptedEngine(["go", "stop"])
+
+    def env_fn(task, history):
+        n_assistant = sum(1 for m in history if m.get("role") == "assistant")
+        done = n_assistant >= 2
+        return f"state-{n_assistant}", float(done), done
+
+    wf = CumulativeWorkflow(eng, env_fn, max_turns=5)
+    ep = asyncio.run(wf.run_with_termination_handling({"q": 1}, "c:0"))
+    traj = ep.trajectories[0]
+    assert len(traj.steps) == 2
+    assert traj.is_cumulative()
+    assert traj.reward == 1.0
+
+
+def test_distillation_workflow_attaches_advantages():
+    from rllm_amd.workflows.builtin import DistillationWorkflow
+
+    student = ScriptedEngine(["answer"])
+
+    class Teacher(ScriptedEngine):
+        @property
+        def supports_token_in_token_out(self):
+            return True
+
+        async def get_token_output_from_token_input(self, token_input, **kw):
+            class Out:
+                logprobs = [-0.05, -0.05]
+
+            return Out()
+
+    wf = DistillationWorkflow(student, Teacher([]), coef=2.0)
+    ep = asyncio.run(wf.run_with_termination_handling({"question": "q"}, "d:0"))
+    step = ep.trajectories[0].steps[0]
+    # student logprobs -0.1 each, teacher -0.05 -> adv = 2*(0.05) = 0.1
+    assert step.advantage == pytest.approx([0.1, 0.1])
