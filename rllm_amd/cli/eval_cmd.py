@@ -26,7 +26,7 @@ def eval_cmd(dataset, agent_ref, evaluator_ref, base_url, model, split, attempts
     from rllm_amd.cli.train import resolve_ref
     from rllm_amd.data.dataset import DatasetRegistry
     from rllm_amd.eval.runner import run_dataset
-    from rllm_amd.gateway.manager import EvalGatewayManager, GatewayManager
+    from rllm_amd.gateway.manager import EvalGatewayManager
 
     registry = DatasetRegistry(registry_dir)
     tasks = registry.load_dataset(dataset, split=split, as_tasks=True)

@@ -6,7 +6,6 @@ Subcommands: train, eval, dataset, view, bench, build-kernels.
 from __future__ import annotations
 
 import importlib
-import json
 import sys
 from pathlib import Path
 

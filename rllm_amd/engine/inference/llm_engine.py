@@ -15,11 +15,10 @@ Design (MI355X-first):
 
 from __future__ import annotations
 
-import math
 import time
 from dataclasses import dataclass, field
 from enum import Enum
-from typing import Any, Callable
+from typing import Callable
 
 import torch
 

@@ -12,7 +12,6 @@ or a test fake) directly — no worker HTTP hop (tinker_adapter.py pattern).
 
 from __future__ import annotations
 
-import asyncio
 import socket
 import threading
 import time

@@ -16,7 +16,6 @@ Behavior preserved:
 from __future__ import annotations
 
 import json
-import time
 from typing import Any, AsyncIterator, Awaitable, Callable
 
 import httpx

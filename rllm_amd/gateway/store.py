@@ -5,7 +5,6 @@ sqlite_store}.py). Memory store for training runs; sqlite for persistence
 from __future__ import annotations
 
 import asyncio
-import json
 import sqlite3
 import threading
 from typing import Protocol

@@ -12,9 +12,8 @@ from __future__ import annotations
 
 import logging
 import shlex
-from typing import Any
 
-from rllm_amd.sandbox.protocol import ExecResult, Sandbox
+from rllm_amd.sandbox.protocol import Sandbox
 from rllm_amd.types import AgentConfig, Task
 
 logger = logging.getLogger(__name__)

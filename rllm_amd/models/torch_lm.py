@@ -5,7 +5,6 @@ part of the MI355X compute path."""
 
 from __future__ import annotations
 
-import math
 
 import torch
 import torch.nn as nn

@@ -17,7 +17,6 @@ from __future__ import annotations
 
 from dataclasses import dataclass, field
 
-import numpy as np
 import torch
 
 from rllm_amd.types import Episode, Step, Trajectory, TrajectoryGroup

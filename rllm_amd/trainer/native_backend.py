@@ -35,7 +35,7 @@ from rllm_amd.trainer.backend_protocol import BackendProtocol
 from rllm_amd.trainer.batch import PackedRow, rows_from_groups
 from rllm_amd.trainer.policy import PolicyTrainer, PolicyTrainerConfig
 from rllm_amd.types import Episode, TrajectoryGroup
-from rllm_amd.utils.tokenizer import ByteTokenizer, load_tokenizer
+from rllm_amd.utils.tokenizer import ByteTokenizer
 
 logger = logging.getLogger(__name__)
 

@@ -8,7 +8,6 @@ from __future__ import annotations
 import logging
 from dataclasses import dataclass
 
-import torch
 
 from rllm_amd.parser.chat_template_parser import ChatTemplateParser
 from rllm_amd.trainer.batch import PackedRow, pack_rows, split_rows_token_balanced
