@@ -202,6 +202,12 @@ class LLMEngine:
             if batch and tokens + n > self.max_num_batched_tokens:
                 break
             need = KVCache.pages_needed(n + 1)
+            if need > self.kv.num_pages - 1:
+                # can NEVER fit, even with the whole pool: fail loudly
+                # instead of spinning (liveness guard)
+                self.waiting.pop(0)
+                self._finish(seq, "error")
+                continue
             if need > self.kv.num_free_pages:
                 break
             seq.pages = self.kv.alloc(need)

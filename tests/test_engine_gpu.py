@@ -197,3 +197,23 @@ def test_kv_preemption_and_resume():
     assert all(len(o.token_ids) == 64 for o in outs)
     assert all(len(o.logprobs) == 64 for o in outs)
     assert engine.kv.num_free_pages == engine.kv.num_pages - 1
+
+
+@requires_gpu
+def test_oversized_prompt_fails_loudly():
+    """A prompt that can never fit the KV pool finishes with an error
+    instead of wedging the scheduler (liveness guard)."""
+    model = tiny_model(seed=41)
+    from rllm_amd.engine.inference.kv_cache import KVCache
+
+    kv = KVCache(model.cfg.num_layers, model.cfg.num_kv_heads, model.cfg.head_dim, 4, device="cuda")
+    engine = LLMEngine(model, kv_cache=kv, eos_token_id=None, seed=1)
+    engine.add_request("big", list(range(1, 200)), SamplingParams(max_tokens=4))
+    engine.add_request("ok", list(range(1, 20)), SamplingParams(max_tokens=4))
+    for _ in range(50):
+        if not engine.has_unfinished():
+            break
+        engine.step()
+    outs = {o.request_id: o for o in engine.pop_finished()}
+    assert outs["big"].finish_reason == "error"
+    assert len(outs["ok"].token_ids) == 4
