@@ -68,6 +68,33 @@ class PolicyTrainerConfig:
         )
 
 
+def tis_weights(old_lp: torch.Tensor, rollout_lp: torch.Tensor, *,
+                mode: str | None, cap: float,
+                seq_ids: torch.Tensor | None = None) -> torch.Tensor | None:
+    """Truncated importance-sampling correction weights
+    (reference verl_backend.py:663-676).
+
+    mode "token": per-token exp(old - rollout), clamped to cap.
+    mode "sequence": one weight per sequence, exp(sum of per-token
+    log-ratios over that sequence), clamped, broadcast back to its tokens
+    (requires seq_ids: for each loss row, which packed sequence it is in).
+    """
+    if mode is None:
+        return None
+    log_ratio = old_lp - rollout_lp
+    if mode == "sequence":
+        if seq_ids is None:
+            raise ValueError("sequence-level TIS needs seq_ids")
+        n_seq = int(seq_ids.max().item()) + 1 if seq_ids.numel() else 0
+        seq_sum = torch.zeros(n_seq, dtype=log_ratio.dtype, device=log_ratio.device)
+        seq_sum.index_add_(0, seq_ids, log_ratio)
+        w_seq = torch.exp(seq_sum).clamp(max=cap)
+        return w_seq[seq_ids]
+    if mode != "token":
+        raise ValueError(f"unknown tis_mode {mode!r}")
+    return torch.exp(log_ratio).clamp(max=cap)
+
+
 class PolicyTrainer:
     def __init__(self, model, ref_model=None, config: PolicyTrainerConfig | None = None):
         self.model = model
@@ -97,14 +124,10 @@ class PolicyTrainer:
         return lp, (ent.mean().item() if ent is not None else None)
 
     # ------------------------------------------------------------------
-    def _tis_weights(self, old_lp: torch.Tensor, rollout_lp: torch.Tensor) -> torch.Tensor | None:
-        """Truncated importance-sampling correction weights
-        (reference verl_backend.py:663-676)."""
-        if self.cfg.tis_mode is None:
-            return None
-        log_ratio = old_lp - rollout_lp
-        w = torch.exp(log_ratio).clamp(max=self.cfg.tis_cap)
-        return w
+    def _tis_weights(self, old_lp: torch.Tensor, rollout_lp: torch.Tensor,
+                     seq_ids: torch.Tensor | None = None) -> torch.Tensor | None:
+        return tis_weights(old_lp, rollout_lp, mode=self.cfg.tis_mode,
+                           cap=self.cfg.tis_cap, seq_ids=seq_ids)
 
     def update_policy(self, rows: list[PackedRow], old_logprob_fn=None) -> dict:
         """One optimizer step over a mini-batch of packed rows.
@@ -161,7 +184,11 @@ class PolicyTrainer:
                     ref_hidden = self.ref_model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
                     ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.ref_model.lm_weight,
                                                     tgt, chunk=cfg.entropy_chunk, want_entropy=False)
-                tis_w = self._tis_weights(old_lp, rollout_lp)
+                seq_ids = None
+                if cfg.tis_mode == "sequence" or cfg.loss_agg_mode == "seq-mean-token-mean":
+                    seq_ids = torch.bucketize(
+                        rows_idx, torch.tensor(batch.cu_seqlens[1:-1], device=rows_idx.device))
+                tis_w = self._tis_weights(old_lp, rollout_lp, seq_ids=seq_ids)
 
             if micro is micros[0]:
                 from rllm_amd.utils.offpolicy import compute_offpolicy_metrics
@@ -178,7 +205,6 @@ class PolicyTrainer:
             elif cfg.loss_agg_mode == "seq-mean-token-mean":
                 # per-sequence token mean, then mean over the global row count
                 n_global_rows = pdist.all_reduce_scalar(float(batch.n_rows), op="sum") if world > 1 else batch.n_rows
-                seq_ids = torch.bucketize(rows_idx, torch.tensor(batch.cu_seqlens[1:-1], device=rows_idx.device))
                 loss = 0.0
                 for s in seq_ids.unique():
                     m = seq_ids == s

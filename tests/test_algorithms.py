@@ -159,3 +159,30 @@ def test_rejection_sampling_min_trajs():
     fg, fe, metrics = apply_rejection_sampling_and_filtering(eps, groups, cfg, state)
     assert fg == []
     assert metrics["batch/groups_dropped_insufficient_trajs"] == 1
+
+
+def test_tis_weights_token_and_sequence():
+    import torch
+
+    from rllm_amd.trainer.policy import tis_weights
+
+    old = torch.tensor([0.0, -1.0, -2.0, -0.5])
+    roll = torch.tensor([-0.5, -1.0, -1.0, -0.5])
+    # token mode: exp(old-roll) clamped
+    w = tis_weights(old, roll, mode="token", cap=1.2)
+    expect = torch.exp(old - roll).clamp(max=1.2)
+    assert torch.allclose(w, expect)
+    # sequence mode: rows 0-1 = seq 0, rows 2-3 = seq 1
+    seq_ids = torch.tensor([0, 0, 1, 1])
+    w = tis_weights(old, roll, mode="sequence", cap=10.0, seq_ids=seq_ids)
+    w0 = torch.exp((old - roll)[:2].sum()).clamp(max=10.0)
+    w1 = torch.exp((old - roll)[2:].sum()).clamp(max=10.0)
+    assert torch.allclose(w, torch.stack([w0, w0, w1, w1]))
+    # disabled / errors
+    assert tis_weights(old, roll, mode=None, cap=1.0) is None
+    import pytest
+
+    with pytest.raises(ValueError):
+        tis_weights(old, roll, mode="sequence", cap=1.0)
+    with pytest.raises(ValueError):
+        tis_weights(old, roll, mode="banana", cap=1.0)
