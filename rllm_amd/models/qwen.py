@@ -116,11 +116,26 @@ class QwenLayer(nn.Module):
         self.gate_up_proj = nn.Parameter(torch.empty(2 * cfg.intermediate_size, H, dtype=torch.bfloat16))
         self.down_proj = nn.Parameter(torch.empty(H, cfg.intermediate_size, dtype=torch.bfloat16))
 
+    # LoRA hooks (models/lora.py): `lora` (a ParameterDict of A/B pairs) is
+    # registered by inject_lora; no class default — a class attribute would
+    # shadow nn.Module submodule lookup. `_lora_on` is stamped per-forward
+    # by QwenModel from its enabled flag.
+    lora_scale: float = 1.0
+    _lora_on: bool = False
+
+    def _lin(self, x: torch.Tensor, name: str, b: torch.Tensor | None = None) -> torch.Tensor:
+        y = _linear(x, getattr(self, name), b)
+        if self._lora_on and f"{name}_A" in self.lora:
+            from rllm_amd.models import lora as _lora
+
+            y = y + _lora.lora_delta(self, name, x)
+        return y
+
     def _qkv(self, hidden: torch.Tensor, positions: torch.Tensor, cos_t, sin_t):
         cfg = self.cfg
         T = hidden.shape[0]
         x = ops.rmsnorm(hidden, self.input_layernorm, cfg.rms_eps)
-        qkv = _linear(x, self.qkv_proj, self.qkv_bias)
+        qkv = self._lin(x, "qkv_proj", self.qkv_bias)
         q, k, v = qkv.split([cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1)
         q = q.view(T, cfg.num_heads, cfg.head_dim)
         k = k.view(T, cfg.num_kv_heads, cfg.head_dim)
@@ -131,9 +146,9 @@ class QwenLayer(nn.Module):
     def _finish(self, hidden: torch.Tensor, attn_out: torch.Tensor) -> torch.Tensor:
         cfg = self.cfg
         T = hidden.shape[0]
-        hidden = hidden + _linear(attn_out.reshape(T, cfg.q_size), self.o_proj)
+        hidden = hidden + self._lin(attn_out.reshape(T, cfg.q_size), "o_proj")
         x = ops.rmsnorm(hidden, self.post_attention_layernorm, cfg.rms_eps)
-        mlp = _linear(ops.swiglu(_linear(x, self.gate_up_proj)), self.down_proj)
+        mlp = self._lin(ops.swiglu(self._lin(x, "gate_up_proj")), "down_proj")
         return hidden + mlp
 
     # -- training path -------------------------------------------------------
@@ -167,15 +182,26 @@ class QwenLayer(nn.Module):
         T = h.shape[0]
         k_pages, v_pages = kv_cache[layer_idx]
         x = ops.add_rmsnorm_(h, delta, self.input_layernorm, cfg.rms_eps)
-        qkv = ops.linear_decode(x, self.qkv_proj)
+        qkv = self._lin_decode(x, "qkv_proj")
         q = ops.qkv_rope_cache(qkv, self.qkv_bias, k_pages, v_pages, cos_t, sin_t,
                                positions, slot_mapping, cfg.num_heads, cfg.num_kv_heads)
         attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
                                 1.0 / math.sqrt(cfg.head_dim))
-        attn_delta = ops.linear_decode(attn.reshape(T, cfg.q_size), self.o_proj)
+        attn_delta = self._lin_decode(attn.reshape(T, cfg.q_size), "o_proj")
         x = ops.add_rmsnorm_(h, attn_delta, self.post_attention_layernorm, cfg.rms_eps)
-        mlp_delta = ops.linear_decode(ops.swiglu(ops.linear_decode(x, self.gate_up_proj)), self.down_proj)
+        mlp_delta = self._lin_decode(ops.swiglu(self._lin_decode(x, "gate_up_proj")), "down_proj")
         return mlp_delta
+
+    def _lin_decode(self, x: torch.Tensor, name: str) -> torch.Tensor:
+        """Decode-path linear: hipBLASLt skinny GEMM + optional unmerged-LoRA
+        term (rollouts normally run MERGED — merge_lora_ — so this branch is
+        off; kept for logprob-exactness checks against the train path)."""
+        y = ops.linear_decode(x, getattr(self, name))
+        if self._lora_on and f"{name}_A" in self.lora:
+            from rllm_amd.models import lora as _lora
+
+            y = y + _lora.lora_delta(self, name, x)
+        return y
 
 
 class QwenModel(nn.Module):
@@ -200,6 +226,16 @@ class QwenModel(nn.Module):
     def lm_weight(self) -> torch.Tensor:
         return self.embed_tokens if self.lm_head is None else self.lm_head
 
+    def _stamp_lora(self) -> None:
+        """Propagate the model-level adapter state to the layers once per
+        forward (merged adapters live inside the base weights, so the
+        low-rank term must be off)."""
+        on = bool(getattr(self, "lora_enabled", False)) and not getattr(self, "lora_merged", False)
+        if on or getattr(self, "_lora_stamped", False):
+            for layer in self.layers:
+                layer._lora_on = on
+            self._lora_stamped = on
+
     def init_random(self, seed: int = 0):
         """Random init at HF-like scales (std 0.02)."""
         gen = torch.Generator(device="cpu").manual_seed(seed)
@@ -223,6 +259,7 @@ class QwenModel(nn.Module):
     def forward_train(self, input_ids: torch.Tensor, positions: torch.Tensor,
                       cu_seqlens: list[int]) -> torch.Tensor:
         """Packed varlen forward -> final hidden states [T, H] (after norm)."""
+        self._stamp_lora()
         hidden = self.embed_tokens[input_ids]
         flash_tiles = None
         if self.use_flash_training_attention:
@@ -251,6 +288,7 @@ class QwenModel(nn.Module):
     # -- rollout --------------------------------------------------------------
     @torch.no_grad()
     def forward_prefill(self, input_ids, positions, tiles, kv_cache, slot_mapping):
+        self._stamp_lora()
         hidden = self.embed_tokens[input_ids]
         for i, layer in enumerate(self.layers):
             hidden = layer.forward_prefill(hidden, positions, self.cos_t, self.sin_t,
@@ -259,6 +297,7 @@ class QwenModel(nn.Module):
 
     @torch.no_grad()
     def forward_decode(self, input_ids, positions, kv_cache, slot_mapping, block_tables, seq_lens):
+        self._stamp_lora()
         h = self.embed_tokens[input_ids].contiguous()
         delta = None
         for i, layer in enumerate(self.layers):

@@ -52,6 +52,9 @@ class PolicyTrainerConfig:
     old_logprob_mode: str = "recompute"
     entropy_chunk: int = 16384
     use_ref: bool = True
+    # LoRA training: the KL reference is the actor's own base weights with
+    # adapters disabled (models/lora.py) — no second model copy.
+    ref_from_lora_base: bool = False
 
     @classmethod
     def from_algorithm_config(cls, a: AlgorithmConfig, **over) -> "PolicyTrainerConfig":
@@ -180,7 +183,14 @@ class PolicyTrainer:
                 else:
                     old_lp = old_logprob_fn(batch, rows_idx)
                 ref_lp = None
-                if cfg.use_ref and self.ref_model is not None:
+                if cfg.use_ref and cfg.ref_from_lora_base:
+                    from rllm_amd.models import lora as _lora
+
+                    with _lora.disabled(self.model):
+                        ref_hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
+                        ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.model.lm_weight,
+                                                        tgt, chunk=cfg.entropy_chunk, want_entropy=False)
+                elif cfg.use_ref and self.ref_model is not None:
                     ref_hidden = self.ref_model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
                     ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.ref_model.lm_weight,
                                                     tgt, chunk=cfg.entropy_chunk, want_entropy=False)
