@@ -58,7 +58,7 @@ def test_lora_grpo_update_trains_only_adapters():
     assert m2["actor/grad_norm"] > 0
     # base weights untouched; adapters moved
     assert torch.equal(model.layers[0].qkv_proj, base_qkv)
-    moved = sum(float(p.abs().sum()) for n, p in model.layers[0].lora.items() if n.endswith("_B"))
+    moved = sum(float(p.detach().abs().sum()) for n, p in model.layers[0].lora.items() if n.endswith("_B"))
     assert moved > 0
     # KL vs base is finite and the ref pass didn't flip adapter state
     assert model.lora_enabled
@@ -84,7 +84,7 @@ def test_lora_merged_decode_matches_train_logits():
 
     T = 18
     ids = torch.randint(0, CFG.vocab_size, (T,), device="cuda")
-    pos = torch.arange(T, device="cuda")
+    pos = torch.arange(T, device="cuda", dtype=torch.int32)
     cu = [0, T]
 
     hidden_train = model.forward_train(ids, pos, cu)
