@@ -195,11 +195,52 @@ class ReverseProxy:
     async def forward_stream(self, worker_url: str, path: str, body: dict,
                              headers: dict | None = None) -> AsyncIterator[bytes]:
         """SSE streaming forward; yields raw lines. Caller buffers parsed
-        chunks for trace assembly."""
+        chunks for trace assembly. With a local_handler (colocated native
+        engine — no HTTP worker) the non-streaming result is re-emitted as
+        OpenAI-style chunk events so streaming clients work unchanged."""
+        if self.local_handler is not None:
+            resp = await self.local_handler({"path": path, "body": {**body, "stream": False}})
+            for raw in synthesize_sse_chunks(resp):
+                yield raw
+            return
         url = worker_url.rstrip("/") + path
         async with self.client.stream("POST", url, json=body, headers=headers or {}) as r:
             async for line in r.aiter_lines():
                 yield (line + "\n").encode()
+
+
+def synthesize_sse_chunks(resp: dict) -> list[bytes]:
+    """Turn a finished chat/completions response into the SSE chunk
+    sequence a streaming worker would have produced: role delta → content
+    delta (carrying the injected token_ids/logprobs/prompt_token_ids so
+    trace assembly still sees them) → finish chunk → [DONE]."""
+
+    def ev(obj: dict) -> bytes:
+        return f"data: {json.dumps(obj)}\n\n".encode()
+
+    base = {"id": resp.get("id"), "object": "chat.completion.chunk",
+            "created": resp.get("created"), "model": resp.get("model")}
+    out: list[bytes] = []
+    for i, ch in enumerate(resp.get("choices") or []):
+        msg = ch.get("message") or {}
+        content = msg.get("content") or ch.get("text") or ""
+        out.append(ev({**base, "choices": [
+            {"index": i, "delta": {"role": msg.get("role", "assistant"), "content": ""}}]}))
+        mid = {"index": i, "delta": {"content": content}}
+        if ch.get("token_ids"):
+            mid["token_ids"] = ch["token_ids"]
+        if ch.get("logprobs") is not None:
+            mid["logprobs"] = ch["logprobs"]
+        chunk = {**base, "choices": [mid]}
+        if resp.get("prompt_token_ids"):
+            chunk["prompt_token_ids"] = resp["prompt_token_ids"]
+        if resp.get("weight_version") is not None:
+            chunk["weight_version"] = resp["weight_version"]
+        out.append(ev(chunk))
+        out.append(ev({**base, "choices": [
+            {"index": i, "delta": {}, "finish_reason": ch.get("finish_reason", "stop")}]}))
+    out.append(b"data: [DONE]\n\n")
+    return out
 
 
 def parse_sse_chunks(lines: list[str]) -> list[dict]:

@@ -204,3 +204,35 @@ def test_cumulative_token_mode():
         client.delete_session("cum:0")
     finally:
         gw.stop()
+
+
+def test_streaming_with_local_handler():
+    """stream=True against a colocated local_handler gateway: synthesized SSE
+    chunks reach the client AND the trace still captures token ids/logprobs."""
+    from rllm_amd.gateway.native_adapter import make_torch_lm_local_handler
+    from rllm_amd.models.torch_lm import TinyTorchLM
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    handler = make_torch_lm_local_handler(TinyTorchLM(seed=0), parser)
+    gw = GatewayManager(GatewayConfig(), local_handler=handler, parser=parser)
+    gw.start()
+    try:
+        client = gw.client()
+        client.create_session("st:0", sampling_params={"max_tokens": 4})
+        url = gw.session_url("st:0") + "/chat/completions"
+        with httpx.stream("POST", url, json={
+                "model": "m", "stream": True,
+                "messages": [{"role": "user", "content": "hi"}]}, timeout=60.0) as r:
+            lines = [ln for ln in r.iter_lines() if ln.startswith("data:")]
+        assert lines[-1].strip() == "data: [DONE]"
+        assert len(lines) >= 4  # role, content, finish, DONE
+        traces = client.get_traces("st:0")
+        assert len(traces) == 1
+        assert len(traces[0].completion_token_ids) == 4
+        assert len(traces[0].logprobs) == 4
+        assert traces[0].prompt_token_ids  # injected fields reached the trace
+        client.delete_session("st:0")
+    finally:
+        gw.stop()
