@@ -150,3 +150,55 @@ def _worker_weight_sync(rank, world, q):
 @pytest.mark.timeout(180)
 def test_separated_weight_sync_world2():
     _run_spawn(_worker_weight_sync, "29615")
+
+
+def _worker_separated_topology(rank, world, q):
+    """Full separated-mode control plane: trainer 0 drives a rollout rank
+    through sync → generate → stop (parallel/separated.py)."""
+    try:
+        pdist = _setup(rank, world)
+
+        from rllm_amd.parallel.separated import (
+            RolloutWorker, SeparatedRolloutClient, SeparatedTopology)
+
+        topo = SeparatedTopology(n_trainers=1, n_rollout=1)
+        flat = torch.full((32,), float(rank))
+
+        if topo.is_rollout:
+            seen_versions = []
+
+            class Eng:
+                paused = False
+                weight_version = 0
+
+                def pause(self):
+                    self.paused = True
+
+                def resume(self):
+                    self.paused = False
+
+            eng = Eng()
+
+            def gen(tasks):
+                seen_versions.append(eng.weight_version)
+                return [{"task": t, "out": t * 2} for t in tasks]
+
+            RolloutWorker(topo, flat, gen, engine=eng).serve()
+            assert flat.eq(0.0).all()          # received trainer weights
+            assert eng.weight_version == 7
+            assert seen_versions == [7]        # generated AFTER the sync
+        else:
+            client = SeparatedRolloutClient(topo, flat)
+            client.sync_weights(version=7)
+            results = client.generate([1, 2, 3])
+            assert [r["out"] for r in results] == [2, 4, 6], results
+            client.stop()
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+@pytest.mark.timeout(180)
+def test_separated_topology_world2():
+    _run_spawn(_worker_separated_topology, "29617")
