@@ -152,13 +152,26 @@ class QwenLayer(nn.Module):
         return hidden + mlp
 
     # -- training path -------------------------------------------------------
-    def forward_train(self, hidden, positions, cu_seqlens, cos_t, sin_t, flash_tiles=None):
+    def forward_train(self, hidden, positions, cu_seqlens, cos_t, sin_t,
+                      flash_tiles=None, sp_group=None):
+        """With sp_group (Ulysses, parallel/ulysses.py): hidden/positions are
+        this rank's token shard, cu_seqlens/flash_tiles describe the FULL
+        packed batch; two all-to-alls around attention give full-sequence
+        visibility per head shard."""
         q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
         scale = 1.0 / math.sqrt(self.cfg.head_dim)
-        if flash_tiles is not None:
-            attn = ops.flash_attention_train(q, k, v, *flash_tiles, scale)
+
+        def attn_fn(qh, kh, vh):
+            if flash_tiles is not None:
+                return ops.flash_attention_train(qh, kh, vh, *flash_tiles, scale)
+            return train_attention(qh, kh, vh, cu_seqlens, scale)
+
+        if sp_group is not None:
+            from rllm_amd.parallel import ulysses
+
+            attn = ulysses.ulysses_attention(q, k, v, attn_fn, sp_group)
         else:
-            attn = train_attention(q, k, v, cu_seqlens, scale)
+            attn = attn_fn(q, k, v)
         return self._finish(hidden, attn)
 
     # -- rollout paths (no grad) ----------------------------------------------
@@ -255,10 +268,16 @@ class QwenModel(nn.Module):
 
     # -- training -------------------------------------------------------------
     use_flash_training_attention: bool = True
+    sp_group = None  # Ulysses SP process group (parallel/ulysses.py); None = off
 
     def forward_train(self, input_ids: torch.Tensor, positions: torch.Tensor,
                       cu_seqlens: list[int]) -> torch.Tensor:
-        """Packed varlen forward -> final hidden states [T, H] (after norm)."""
+        """Packed varlen forward -> final hidden states [T, H] (after norm).
+
+        Ulysses mode (self.sp_group set): input_ids/positions are this
+        rank's token shard (parallel/ulysses.py shard_slice), cu_seqlens is
+        the FULL packed batch; the returned hidden covers the local shard.
+        """
         self._stamp_lora()
         hidden = self.embed_tokens[input_ids]
         flash_tiles = None
@@ -269,10 +288,12 @@ class QwenModel(nn.Module):
             if self.gradient_checkpointing and torch.is_grad_enabled():
                 hidden = torch.utils.checkpoint.checkpoint(
                     layer.forward_train, hidden, positions, cu_seqlens,
-                    self.cos_t, self.sin_t, flash_tiles, use_reentrant=False)
+                    self.cos_t, self.sin_t, flash_tiles, self.sp_group,
+                    use_reentrant=False)
             else:
                 hidden = layer.forward_train(hidden, positions, cu_seqlens,
-                                             self.cos_t, self.sin_t, flash_tiles)
+                                             self.cos_t, self.sin_t, flash_tiles,
+                                             self.sp_group)
         return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
 
     def logprobs_for_tokens(self, input_ids, positions, cu_seqlens, targets,
