@@ -363,6 +363,10 @@ class LLMEngine:
         key = (Bp, temp)
         if key in self._graphs:
             return self._graphs[key]
+        # tune the hipblaslt algo for THIS padded batch's GEMM shapes before
+        # capture — the default heuristic has a ~20 us/call floor at skinny M
+        ops.pretune_decode_shapes(ops.decode_gemm_shapes(self.cfg, [Bp]),
+                                  device=self.device)
         # warm up eager once (cuBLAS-style workspace init), then capture
         torch.cuda.synchronize()
         self._decode_compute(Bp, temp)

@@ -307,13 +307,66 @@ _USE_SKINNY = _os.environ.get("RLLM_SKINNY_GEMM", "0") == "1"
 
 
 def linear_decode(x, w, bias=None):
-    """Decode-path linear. The in-repo weight-streaming skinny kernel is
-    correct but currently ~3x slower than hipBLASLt at these shapes
-    (16-col tiles over-read A from L2; redesign tracked for round 2) —
-    opt in with RLLM_SKINNY_GEMM=1."""
+    """Decode-path linear. Default: hipBLASLt with per-shape TUNED algo
+    selection (hbl_tuned.hip) when `pretune_decode_shapes` has run for this
+    (M,N,K) — torch.matmul's default heuristic leaves a ~19-25 us/call
+    floor at M<=512 on MI355X. Falls back to the default heuristic for
+    untuned shapes. The in-repo weight-streaming skinny kernel is correct
+    but slower (RLLM_SKINNY_GEMM=1 to opt in; redesign notes in TODO)."""
     if _USE_SKINNY and x.shape[0] <= 512 and x.shape[-1] % 32 == 0:
         return require_ext().skinny_gemm(x, w, bias)
+    if bias is None and _TUNED_SHAPES and (x.shape[0], w.shape[0], w.shape[1]) in _TUNED_SHAPES:
+        return require_ext().hbl_mm(x, w)
     return torch.nn.functional.linear(x, w, bias)
+
+
+_TUNED_SHAPES: set = set()
+_TUNE_DISABLED = _os.environ.get("RLLM_TUNED_GEMM", "1") == "0"
+
+
+def pretune_decode_shapes(shapes, device="cuda", iters: int = 20, verbose: bool = True):
+    """Sweep hipblaslt's heuristic candidates once per (M, N, K) decode
+    shape and cache the fastest valid algo (hbl_tuned.hip). Call from
+    engine init BEFORE any hipGraph capture; linear_decode then uses the
+    tuned algo for exactly these shapes. Returns {shape: best_us}."""
+    if _TUNE_DISABLED:
+        return {}
+    C = require_ext()
+    out = {}
+    for (M, N, K) in shapes:
+        key = (int(M), int(N), int(K))
+        if key in _TUNED_SHAPES:
+            continue
+        x = torch.randn(M, K, device=device).to(torch.bfloat16)
+        w = torch.randn(N, K, device=device).to(torch.bfloat16)
+        try:
+            best_us, n_cand, n_valid, idx = C.hbl_tune(x, w, iters)
+        except RuntimeError as e:  # no valid algo — keep default path
+            if verbose:
+                print(f"[ops.pretune] {key}: tune failed ({e}); using default heuristic")
+            continue
+        _TUNED_SHAPES.add(key)
+        out[key] = best_us
+        if verbose:
+            print(f"[ops.pretune] {key}: {best_us:.2f} us "
+                  f"(algo {int(idx)}, {int(n_valid)}/{int(n_cand)} valid)")
+    return out
+
+
+def decode_gemm_shapes(cfg, batch_sizes) -> list:
+    """The (M, N, K) set the fused decode path hits for a model config:
+    qkv, o, gate_up, down projections (+ lm head for logits) at each padded
+    graph batch size."""
+    shapes = []
+    for M in batch_sizes:
+        shapes += [
+            (M, cfg.q_size + 2 * cfg.kv_size, cfg.hidden_size),
+            (M, cfg.hidden_size, cfg.q_size),
+            (M, 2 * cfg.intermediate_size, cfg.hidden_size),
+            (M, cfg.hidden_size, cfg.intermediate_size),
+            (M, cfg.vocab_size, cfg.hidden_size),
+        ]
+    return shapes
 
 
 def sample_logprob(logits, temperature: float, seed: int, step: int, step_tensor=None):

@@ -26,6 +26,7 @@ SOURCES = [
     "adamw.hip",
     "sampling.hip",
     "skinny_gemm.hip",
+    "hbl_tuned.hip",
     "attention.hip",
     "bindings.cpp",
 ]
@@ -95,7 +96,7 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     link_cmd = (
         ["hipcc", "-shared", "-o", str(SO_PATH)]
         + objs
-        + [f"-L{lib_dir}", "-ltorch", "-ltorch_python", "-lc10", "-ltorch_hip", "-lc10_hip", "-lamdhip64",
+        + [f"-L{lib_dir}", "-ltorch", "-ltorch_python", "-lc10", "-ltorch_hip", "-lc10_hip", "-lamdhip64", "-L/opt/rocm/lib", "-lhipblaslt",
            f"-Wl,-rpath,{lib_dir}"]
     )
     if verbose:

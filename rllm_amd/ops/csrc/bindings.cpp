@@ -56,6 +56,11 @@ torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bia
 // skinny_gemm.hip
 torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, c10::optional<torch::Tensor> bias);
 
+// hbl_tuned.hip
+std::vector<double> hbl_tune(torch::Tensor x, torch::Tensor w, int64_t iters);
+bool hbl_has(int64_t M, int64_t N, int64_t K);
+torch::Tensor hbl_mm(torch::Tensor x, torch::Tensor w);
+
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
                                           int64_t seed, int64_t step,
@@ -81,6 +86,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");
   m.def("add_rmsnorm_", &add_rmsnorm_, "Fused residual add (in-place) + RMSNorm");
   m.def("skinny_gemm", &skinny_gemm, "Weight-streaming skinny-M GEMM (decode path)");
+  m.def("hbl_tune", &hbl_tune, "Sweep hipblaslt heuristic algos for a decode shape; cache winner");
+  m.def("hbl_has", &hbl_has, "Is this (M,N,K) tuned?");
+  m.def("hbl_mm", &hbl_mm, "Tuned hipblaslt GEMM (graph-capture safe)");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

@@ -1,0 +1,189 @@
+// Tuned hipBLASLt path for decode-shape GEMMs (y[M,N] = x[M,K] @ W[N,K]^T).
+//
+// torch.matmul's default hipBLASLt heuristic shows a ~19-25 us floor at
+// M<=512 decode shapes on MI355X where the weight-stream bound is 1-9 us.
+// This module sweeps the heuristic's candidate list ONCE per shape (outputs
+// validated against the default path, timed with hip events), caches the
+// winning algo, and replays it for every later call — including inside
+// hipGraph capture (workspace is preallocated; no per-call allocations
+// beyond the output tensor).
+//
+// This is deliberate use of the vendor GEMM library (allowed for plain
+// GEMMs) with explicit algorithm selection instead of the default
+// heuristic; no TunableOp (its full sweep has crashed boxes — here only
+// heuristic-advertised algos for this exact problem are tried, each
+// validated before timing).
+
+#include "common.hpp"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <map>
+#include <mutex>
+#include <tuple>
+#include <vector>
+
+#define HBL_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipblasStatus_t _s = (expr);                                               \
+    TORCH_CHECK(_s == HIPBLAS_STATUS_SUCCESS, "hipblaslt error ", (int)_s,     \
+                " at ", #expr);                                                \
+  } while (0)
+
+namespace {
+
+struct ShapeKey {
+  int M, N, K;
+  bool operator<(const ShapeKey& o) const {
+    return std::tie(M, N, K) < std::tie(o.M, o.N, o.K);
+  }
+};
+
+struct CachedMatmul {
+  hipblasLtMatmulDesc_t desc{};
+  hipblasLtMatrixLayout_t la{}, lb{}, lc{};
+  hipblasLtMatmulAlgo_t algo{};
+  bool valid = false;
+};
+
+hipblasLtHandle_t g_handle = nullptr;
+void* g_workspace = nullptr;
+constexpr size_t kWorkspaceBytes = 64u << 20;
+std::map<ShapeKey, CachedMatmul> g_cache;
+std::mutex g_mu;
+
+hipblasLtHandle_t handle() {
+  if (!g_handle) HBL_CHECK(hipblasLtCreate(&g_handle));
+  if (!g_workspace) {
+    hipError_t e = hipMalloc(&g_workspace, kWorkspaceBytes);
+    TORCH_CHECK(e == hipSuccess, "hipblaslt workspace alloc failed");
+  }
+  return g_handle;
+}
+
+// Row-major y[M,N] = x[M,K] @ W[N,K]^T as column-major
+// D[N,M] = A^T(W:[K,N],ld K) @ B(x:[K,M],ld K), C=D ld N.
+CachedMatmul make_desc(int M, int N, int K) {
+  CachedMatmul cm;
+  HBL_CHECK(hipblasLtMatmulDescCreate(&cm.desc, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  int32_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
+  HBL_CHECK(hipblasLtMatmulDescSetAttribute(cm.desc, HIPBLASLT_MATMUL_DESC_TRANSA,
+                                            &opT, sizeof(opT)));
+  HBL_CHECK(hipblasLtMatmulDescSetAttribute(cm.desc, HIPBLASLT_MATMUL_DESC_TRANSB,
+                                            &opN, sizeof(opN)));
+  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.la, HIP_R_16BF, K, N, K));
+  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.lb, HIP_R_16BF, K, M, K));
+  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.lc, HIP_R_16BF, N, M, N));
+  return cm;
+}
+
+void run_matmul(const CachedMatmul& cm, const void* x, const void* w, void* y,
+                const hipblasLtMatmulAlgo_t* algo, hipStream_t stream) {
+  float alpha = 1.f, beta = 0.f;
+  HBL_CHECK(hipblasLtMatmul(handle(), cm.desc, &alpha,
+                            w, cm.la, x, cm.lb, &beta,
+                            y, cm.lc, y, cm.lc,
+                            algo, g_workspace, kWorkspaceBytes, stream));
+}
+
+}  // namespace
+
+// Sweep heuristic candidates for this shape; cache the fastest valid algo.
+// Returns {best_us, n_candidates, n_valid, best_index}.
+std::vector<double> hbl_tune(torch::Tensor x, torch::Tensor w, int64_t iters) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dtype() == torch::kBFloat16 && w.is_contiguous());
+  const int M = (int)x.size(0), K = (int)x.size(1), N = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == K);
+  hipStream_t stream = at::hip::getCurrentHIPStream().stream();
+
+  std::lock_guard<std::mutex> lk(g_mu);
+  CachedMatmul cm = make_desc(M, N, K);
+
+  hipblasLtMatmulPreference_t pref;
+  HBL_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  uint64_t ws = kWorkspaceBytes;
+  HBL_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
+
+  constexpr int kTopN = 48;
+  hipblasLtMatmulHeuristicResult_t results[kTopN];
+  int returned = 0;
+  HBL_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle(), cm.desc, cm.la, cm.lb,
+                                            cm.lc, cm.lc, pref, kTopN, results,
+                                            &returned));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  TORCH_CHECK(returned > 0, "no hipblaslt algos for shape ", M, "x", N, "x", K);
+
+  // reference from the default torch path (also bf16 -> loose tolerance)
+  auto ref = torch::matmul(x, w.t()).contiguous();
+  const double tol = ref.abs().max().item<double>() * 0.05 + 0.1;
+
+  auto y = torch::empty({M, N}, x.options());
+  hipEvent_t ev0, ev1;
+  (void)hipEventCreate(&ev0);
+  (void)hipEventCreate(&ev1);
+
+  double best_us = 1e30;
+  int best_idx = -1, n_valid = 0;
+  for (int i = 0; i < returned; ++i) {
+    if (results[i].state != HIPBLAS_STATUS_SUCCESS) continue;
+    // probe + validate
+    float alpha = 1.f, beta = 0.f;
+    hipblasStatus_t st = hipblasLtMatmul(handle(), cm.desc, &alpha,
+                                         w.data_ptr(), cm.la, x.data_ptr(), cm.lb,
+                                         &beta, y.data_ptr(), cm.lc,
+                                         y.data_ptr(), cm.lc, &results[i].algo,
+                                         g_workspace, kWorkspaceBytes, stream);
+    if (st != HIPBLAS_STATUS_SUCCESS) continue;
+    double err = (y.to(torch::kFloat32) - ref.to(torch::kFloat32)).abs().max().item<double>();
+    if (!(err <= tol)) continue;
+    ++n_valid;
+    // warm
+    for (int r = 0; r < 3; ++r)
+      run_matmul(cm, x.data_ptr(), w.data_ptr(), y.data_ptr(), &results[i].algo, stream);
+    (void)hipEventRecord(ev0, stream);
+    for (int64_t r = 0; r < iters; ++r)
+      run_matmul(cm, x.data_ptr(), w.data_ptr(), y.data_ptr(), &results[i].algo, stream);
+    (void)hipEventRecord(ev1, stream);
+    (void)hipEventSynchronize(ev1);
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, ev0, ev1);
+    double us = ms * 1000.0 / (double)iters;
+    if (us < best_us) {
+      best_us = us;
+      best_idx = i;
+    }
+  }
+  (void)hipEventDestroy(ev0);
+  (void)hipEventDestroy(ev1);
+  TORCH_CHECK(best_idx >= 0, "no valid hipblaslt algo for ", M, "x", N, "x", K);
+
+  cm.algo = results[best_idx].algo;
+  cm.valid = true;
+  g_cache[ShapeKey{M, N, K}] = cm;
+  return {best_us, (double)returned, (double)n_valid, (double)best_idx};
+}
+
+bool hbl_has(int64_t M, int64_t N, int64_t K) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_cache.find(ShapeKey{(int)M, (int)N, (int)K});
+  return it != g_cache.end() && it->second.valid;
+}
+
+torch::Tensor hbl_mm(torch::Tensor x, torch::Tensor w) {
+  const int M = (int)x.size(0), K = (int)x.size(1), N = (int)w.size(0);
+  const CachedMatmul* cm;
+  {
+    std::lock_guard<std::mutex> lk(g_mu);
+    auto it = g_cache.find(ShapeKey{M, N, K});
+    TORCH_CHECK(it != g_cache.end() && it->second.valid,
+                "hbl_mm: shape ", M, "x", N, "x", K, " not tuned (call hbl_tune first)");
+    cm = &it->second;
+  }
+  auto y = torch::empty({M, N}, x.options());
+  run_matmul(*cm, x.data_ptr(), w.data_ptr(), y.data_ptr(), &cm->algo,
+             at::hip::getCurrentHIPStream().stream());
+  return y;
+}
