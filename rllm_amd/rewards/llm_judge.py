@@ -115,3 +115,22 @@ class LLMJudgeReward:
         if not candidate:
             return 0.0
         return self.grade(question, str(reference), candidate)
+
+
+_default_judge: LLMJudgeReward | None = None
+
+
+def llm_judge_reward_fn(task, episode) -> float:
+    """Registry entry ("llm_judge"): endpoint from RLLM_JUDGE_URL /
+    RLLM_JUDGE_MODEL env vars (e.g. the gateway's base url)."""
+    global _default_judge
+    if _default_judge is None:
+        import os
+
+        url = os.environ.get("RLLM_JUDGE_URL")
+        if not url:
+            raise RuntimeError("llm_judge reward needs RLLM_JUDGE_URL "
+                               "(an OpenAI-compatible /v1 base url)")
+        _default_judge = LLMJudgeReward(
+            base_url=url, model=os.environ.get("RLLM_JUDGE_MODEL", "judge"))
+    return _default_judge(task, episode)

@@ -155,4 +155,5 @@ REWARD_FN_REGISTRY = {
     "mcq": "rllm_amd.rewards.reward_fns:mcq_reward_fn",
     "f1": "rllm_amd.rewards.reward_fns:f1_reward_fn",
     "countdown": "rllm_amd.rewards.reward_fns:countdown_reward_fn",
+    "llm_judge": "rllm_amd.rewards.llm_judge:llm_judge_reward_fn",
 }
