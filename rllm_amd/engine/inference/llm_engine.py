@@ -144,6 +144,11 @@ class LLMEngine:
     # ------------------------------------------------------------------
     def add_request(self, request_id: str, prompt_ids: list[int], params: SamplingParams) -> None:
         seq = Sequence(request_id, prompt_ids, params, weight_version=self.weight_version)
+        if len(prompt_ids) >= self.max_model_len:
+            # MAX_PROMPT_LENGTH_EXCEEDED: fail loudly instead of letting the
+            # sequence outgrow the block-table width mid-decode
+            self._finish(seq, "error")
+            return
         self.waiting.append(seq)
 
     def abort(self, request_id: str) -> None:
@@ -467,6 +472,9 @@ class LLMEngine:
         if tok in stop_ids:
             self._finish(seq, "stop")
         elif len(seq.output_ids) >= seq.params.max_tokens:
+            self._finish(seq, "length")
+        elif seq.total_len >= self.max_model_len:
+            # context-window cap: stop before the block table would overflow
             self._finish(seq, "length")
 
     def _finish(self, seq: Sequence, reason: str) -> None:

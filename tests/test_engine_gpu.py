@@ -217,3 +217,25 @@ def test_oversized_prompt_fails_loudly():
     outs = {o.request_id: o for o in engine.pop_finished()}
     assert outs["big"].finish_reason == "error"
     assert len(outs["ok"].token_ids) == 4
+
+
+@requires_gpu
+def test_max_model_len_enforced():
+    """Prompts >= max_model_len are rejected at submit with finish 'error';
+    generation is capped at the context window with finish 'length'."""
+    model = tiny_model(seed=42)
+    from rllm_amd.engine.inference.kv_cache import KVCache
+
+    kv = KVCache(model.cfg.num_layers, model.cfg.num_kv_heads, model.cfg.head_dim, 64, device="cuda")
+    engine = LLMEngine(model, kv_cache=kv, eos_token_id=None, seed=1, max_model_len=48)
+    engine.add_request("toolong", list(range(1, 50)), SamplingParams(max_tokens=4))
+    engine.add_request("capped", list(range(1, 41)), SamplingParams(max_tokens=64))
+    for _ in range(80):
+        if not engine.has_unfinished():
+            break
+        engine.step()
+    outs = {o.request_id: o for o in engine.pop_finished()}
+    assert outs["toolong"].finish_reason == "error"
+    assert outs["capped"].finish_reason == "length"
+    # stopped at the window: prompt 40 + completion <= 48 (+1 step slack)
+    assert len(outs["capped"].token_ids) <= 9
