@@ -59,6 +59,7 @@ class NativeBackend(BackendProtocol):
         n_parallel_tasks: int = 128,
         checkpoint_path: str | None = None,
         seed: int = 0,
+        lora=None,  # models.lora.LoRAConfig | True for defaults | None = full finetune
     ):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
@@ -76,6 +77,7 @@ class NativeBackend(BackendProtocol):
         self.rollout_sampling_params = rollout_sampling_params or {"temperature": 1.0, "top_p": 1.0}
         self.n_parallel_tasks = n_parallel_tasks
         self.checkpoint_path = checkpoint_path
+        self.lora_config = lora
 
         self.model: QwenModel | None = None
         self.ref_model: QwenModel | None = None
@@ -103,7 +105,16 @@ class NativeBackend(BackendProtocol):
         if self.checkpoint_path:
             sd = torch.load(self.checkpoint_path, weights_only=True, map_location=self.device)
             self.model.load_state_dict(sd)
-        if self.use_ref:
+        if self.lora_config is not None:
+            # LoRA: adapters on the actor; the KL reference is the frozen
+            # base with adapters disabled — no second weight copy.
+            from rllm_amd.models import lora as _lora
+
+            cfg = self.lora_config if not isinstance(self.lora_config, bool) else _lora.LoRAConfig()
+            _lora.inject_lora(self.model, cfg)
+            if self.use_ref:
+                self.policy_config.ref_from_lora_base = True
+        elif self.use_ref:
             self.ref_model = QwenModel(self.cfg, device=self.device).init_random(seed=self.seed)
             self.ref_model.load_state_dict(self.model.state_dict())
             for p in self.ref_model.parameters():
