@@ -258,6 +258,9 @@ class PolicyTrainer:
 
     def load_state_dict(self, sd: dict):
         self.model.load_state_dict(sd["model"])
-        self.flat_param.copy_(torch.cat([p.data.reshape(-1) for p in self.model.parameters()]))
+        # only trainable params live in the flat buffer (all of them for a
+        # full finetune; adapters only under LoRA)
+        self.flat_param.copy_(torch.cat(
+            [p.data.reshape(-1) for p in self.model.parameters() if p.requires_grad]))
         self.optim.load_state_dict(sd["optim"])
         self.weight_version = int(sd.get("weight_version", 0))
