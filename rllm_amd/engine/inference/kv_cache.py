@@ -61,3 +61,135 @@ class KVCache:
         per_page = cls.bytes_per_page(num_layers, num_kv_heads, head_dim)
         num_pages = max(2, budget_bytes // per_page)
         return cls(num_layers, num_kv_heads, head_dim, num_pages, device=device)
+
+
+# ---------------------------------------------------------------------------
+# Prefix cache (automatic prefix reuse across requests; the role vLLM's
+# automatic-prefix-caching plays for the reference's multi-turn rollouts —
+# cumulative sessions share page-aligned prefixes, so turn N+1 reuses turn
+# N's pages instead of recomputing prompt KV)
+# ---------------------------------------------------------------------------
+
+
+def _page_hash(parent: int, tokens: tuple) -> int:
+    import xxhash
+
+    h = xxhash.xxh3_64()
+    h.update(parent.to_bytes(8, "little"))
+    for t in tokens:
+        h.update(int(t).to_bytes(4, "little", signed=True))
+    return h.intdigest()
+
+
+class PrefixCache:
+    """Content-addressed full pages layered over a KVCache.
+
+    Pages are keyed by the hash CHAIN of their token content (position-
+    dependent by construction). Shared pages are reference-counted; pages
+    whose refcount drops to zero stay cached (evictable, LRU) until the
+    allocator needs them. `bump_version` (weight sync) invalidates
+    everything — cached KV from old weights is wrong.
+    """
+
+    def __init__(self, kv: KVCache):
+        from collections import OrderedDict
+
+        self.kv = kv
+        self.map: dict[int, int] = {}      # hash -> page
+        self.page_key: dict[int, int] = {} # page -> hash
+        self.rc: dict[int, int] = {}       # page -> refcount (shared pages only)
+        self.lru: "OrderedDict[int, None]" = OrderedDict()  # rc==0 cached pages
+        self.hits = 0
+        self.misses = 0
+
+    # -- hashing ------------------------------------------------------------
+    @staticmethod
+    def page_hashes(tokens: list[int]) -> list[int]:
+        """Hash per FULL page of the token prefix."""
+        out = []
+        h = 0
+        for i in range(0, len(tokens) - PAGE_SIZE + 1, PAGE_SIZE):
+            h = _page_hash(h, tuple(tokens[i : i + PAGE_SIZE]))
+            out.append(h)
+        return out
+
+    # -- lookup / publish ---------------------------------------------------
+    def match(self, tokens: list[int]) -> list[int]:
+        """Longest cached page run for this token prefix; acquires a
+        reference on each returned page."""
+        pages = []
+        for h in self.page_hashes(tokens):
+            page = self.map.get(h)
+            if page is None:
+                break
+            pages.append(page)
+        for p in pages:
+            self.rc[p] = self.rc.get(p, 0) + 1
+            self.lru.pop(p, None)
+        self.hits += len(pages)
+        self.misses += max(0, (len(tokens) // PAGE_SIZE) - len(pages))
+        return pages
+
+    def publish(self, tokens: list[int], pages: list[int], n_owned_prefix: int) -> None:
+        """Register this sequence's full pages under their content hashes.
+        The first n_owned_prefix pages were acquired from the cache (already
+        registered); later pages were allocated privately by the sequence
+        and become shared (the caller must stop treating them as private:
+        release() them instead of kv.free)."""
+        hashes = self.page_hashes(tokens)
+        for i, h in enumerate(hashes):
+            if i < len(pages) and h not in self.map and i >= n_owned_prefix:
+                page = pages[i]
+                self.map[h] = page
+                self.page_key[page] = h
+                # the private owner's implicit reference becomes a counted one
+                self.rc[page] = self.rc.get(page, 0) + 1
+
+    # -- refcounting --------------------------------------------------------
+    def release(self, pages: list[int]) -> None:
+        """Drop one reference per page. Shared pages at rc==0 become
+        evictable; never-shared pages go straight back to the allocator."""
+        for p in pages:
+            if p in self.page_key or p in self.rc:
+                n = self.rc.get(p, 1) - 1
+                if n > 0:
+                    self.rc[p] = n
+                elif p in self.page_key:
+                    self.rc[p] = 0
+                    self.lru[p] = None  # cached, evictable
+                else:
+                    self.rc.pop(p, None)
+                    self.kv.free([p])
+            else:
+                self.kv.free([p])
+
+    def evict(self, n: int) -> int:
+        """Free up to n evictable cached pages back to the allocator."""
+        freed = 0
+        while freed < n and self.lru:
+            page, _ = self.lru.popitem(last=False)
+            key = self.page_key.pop(page, None)
+            if key is not None:
+                self.map.pop(key, None)
+            self.rc.pop(page, None)
+            self.kv.free([page])
+            freed += 1
+        return freed
+
+    def clear(self) -> None:
+        """Weight version bumped: all cached KV is stale."""
+        for page in list(self.lru):
+            key = self.page_key.pop(page, None)
+            if key is not None:
+                self.map.pop(key, None)
+            self.rc.pop(page, None)
+            self.kv.free([page])
+        self.lru.clear()
+        # pages still referenced by live sequences keep their KV (those
+        # sequences continue decoding on it) but must not be re-matched:
+        for page in list(self.page_key):
+            self.map.pop(self.page_key.pop(page), None)
+
+    @property
+    def evictable(self) -> int:
+        return len(self.lru)

@@ -23,8 +23,8 @@ from typing import Callable
 import torch
 
 from rllm_amd import ops
-from rllm_amd.engine.inference.kv_cache import PAGE_SIZE, KVCache
-from rllm_amd.models.qwen import QwenModel, make_prefill_tiles
+from rllm_amd.engine.inference.kv_cache import PAGE_SIZE, KVCache, PrefixCache
+from rllm_amd.models.qwen import QwenModel, make_prefill_tiles, make_prefill_tiles_cached
 
 
 @dataclass
@@ -60,7 +60,7 @@ class SeqState(Enum):
 class Sequence:
     __slots__ = ("request_id", "prompt_ids", "output_ids", "logprobs", "params",
                  "state", "pages", "finish_reason", "weight_version", "arrival_time",
-                 "first_token_time", "step_counter")
+                 "first_token_time", "step_counter", "cached_len", "n_cached_pages")
 
     def __init__(self, request_id: str, prompt_ids: list[int], params: SamplingParams,
                  weight_version: int = 0):
@@ -76,6 +76,8 @@ class Sequence:
         self.arrival_time = time.monotonic()
         self.first_token_time: float | None = None
         self.step_counter = 0
+        self.cached_len = 0        # prefix-cached tokens at prefill admission
+        self.n_cached_pages = 0    # leading shared pages in self.pages
 
     @property
     def total_len(self) -> int:
@@ -104,7 +106,8 @@ class LLMEngine:
     def __init__(self, model: QwenModel, kv_cache: KVCache | None = None,
                  max_num_seqs: int = 1024, max_num_batched_tokens: int = 8192,
                  kv_budget_bytes: int | None = None, eos_token_id: int | None = None,
-                 seed: int = 0, use_hip_graph: bool = True, max_model_len: int = 8192):
+                 seed: int = 0, use_hip_graph: bool = True, max_model_len: int = 8192,
+                 enable_prefix_caching: bool = True):
         self.model = model
         self.cfg = model.cfg
         self.device = next(model.parameters()).device
@@ -120,8 +123,11 @@ class LLMEngine:
         self.max_num_batched_tokens = max_num_batched_tokens
         self.eos_token_id = eos_token_id
         self.seed = seed
-        self.weight_version = 0
+        self._weight_version = 0
         self._paused = False
+        # automatic prefix caching: cross-request page reuse keyed on token
+        # content (multi-turn/cumulative rollouts share prompt prefixes)
+        self.prefix_cache = PrefixCache(self.kv) if enable_prefix_caching else None
 
         self.waiting: list[Sequence] = []
         self.running: list[Sequence] = []
@@ -138,6 +144,35 @@ class LLMEngine:
         self._graph_pool = None
         self._gb = None                # static device buffers dict
         self._hb = None                # pinned host staging dict
+
+    # ------------------------------------------------------------------
+    @property
+    def weight_version(self) -> int:
+        return self._weight_version
+
+    @weight_version.setter
+    def weight_version(self, v: int) -> None:
+        if v != self._weight_version and self.prefix_cache is not None:
+            # cached KV was computed under the old weights — stale
+            self.prefix_cache.clear()
+        self._weight_version = v
+
+    def _alloc_pages(self, n: int) -> list[int]:
+        """Page allocation that evicts idle prefix-cache pages under pressure."""
+        short = n - self.kv.num_free_pages
+        if short > 0 and self.prefix_cache is not None:
+            self.prefix_cache.evict(short)
+        return self.kv.alloc(n)
+
+    def _release_pages(self, pages: list[int]) -> None:
+        if self.prefix_cache is not None:
+            self.prefix_cache.release(pages)
+        else:
+            self.kv.free(pages)
+
+    def _effective_free_pages(self) -> int:
+        extra = self.prefix_cache.evictable if self.prefix_cache is not None else 0
+        return self.kv.num_free_pages + extra
 
     # ------------------------------------------------------------------
     # Request lifecycle
@@ -201,11 +236,10 @@ class LLMEngine:
             return []
         batch: list[Sequence] = []
         tokens = 0
+        pc = self.prefix_cache
         while self.waiting and len(self.running) + len(batch) < self.max_num_seqs:
             seq = self.waiting[0]
             n = seq.total_len
-            if batch and tokens + n > self.max_num_batched_tokens:
-                break
             need = KVCache.pages_needed(n + 1)
             if need > self.kv.num_pages - 1:
                 # can NEVER fit, even with the whole pool: fail loudly
@@ -213,17 +247,45 @@ class LLMEngine:
                 self.waiting.pop(0)
                 self._finish(seq, "error")
                 continue
-            if need > self.kv.num_free_pages:
+            # prefix cache: reuse pages whose token content matches
+            matched: list[int] = []
+            if pc is not None:
+                matched = pc.match(seq.prompt_ids + seq.output_ids)
+            # always leave >=1 token to prefill (we need its logits)
+            cached_len = min(PAGE_SIZE * len(matched), n - 1)
+            suffix = n - cached_len
+            if batch and tokens + suffix > self.max_num_batched_tokens:
+                if matched:
+                    pc.release(matched)
                 break
-            seq.pages = self.kv.alloc(need)
+            if need - len(matched) > self._effective_free_pages():
+                if matched:
+                    pc.release(matched)
+                break
+            seq.pages = matched + self._alloc_pages(need - len(matched))
+            seq.cached_len = cached_len
+            seq.n_cached_pages = len(matched)
             batch.append(self.waiting.pop(0))
-            tokens += n
+            tokens += suffix
         return batch
 
     def _slot(self, seq: Sequence, pos: int) -> int:
         return seq.pages[pos // PAGE_SIZE] * PAGE_SIZE + pos % PAGE_SIZE
 
     def _run_prefill(self, batch: list[Sequence]) -> int:
+        if any(seq.cached_len > 0 for seq in batch):
+            n_tok = self._run_prefill_cached(batch)
+        else:
+            n_tok = self._run_prefill_full(batch)
+        for seq in batch:
+            if seq.state != SeqState.FINISHED:
+                seq.state = SeqState.RUNNING
+                self.running.append(seq)
+            if seq.first_token_time is None:
+                seq.first_token_time = time.monotonic()
+        return n_tok
+
+    def _run_prefill_full(self, batch: list[Sequence]) -> int:
         device = self.device
         input_ids, positions, slot_mapping, seqlens, last_rows = [], [], [], [], []
         row = 0
@@ -245,12 +307,41 @@ class LLMEngine:
         hidden = self.model.forward_prefill(ids_t, pos_t, tiles, self.kv, slots_t)
         last_hidden = hidden[torch.tensor(last_rows, device=device, dtype=torch.long)]
         self._sample_and_append(batch, last_hidden)
+        return len(input_ids)
+
+    def _run_prefill_cached(self, batch: list[Sequence]) -> int:
+        """Suffix-only prefill: cached prefix K/V come from shared pages;
+        only tokens past each sequence's cached_len run the forward."""
+        device = self.device
+        input_ids, positions, slot_suffix, slot_full = [], [], [], []
+        full_lens, cached_lens, suffix_rows, last_rows = [], [], [], []
+        full_row = 0
+        suffix_row = 0
         for seq in batch:
-            if seq.state != SeqState.FINISHED:
-                seq.state = SeqState.RUNNING
-                self.running.append(seq)
-            if seq.first_token_time is None:
-                seq.first_token_time = time.monotonic()
+            toks = seq.prompt_ids + seq.output_ids
+            n, c = len(toks), seq.cached_len
+            input_ids.extend(toks[c:])
+            positions.extend(range(c, n))
+            slot_suffix.extend(self._slot(seq, p) for p in range(c, n))
+            slot_full.extend(self._slot(seq, p) for p in range(n))
+            suffix_rows.extend(range(full_row + c, full_row + n))
+            full_lens.append(n)
+            cached_lens.append(c)
+            full_row += n
+            suffix_row += n - c
+            last_rows.append(suffix_row - 1)
+
+        ids_t = torch.tensor(input_ids, device=device, dtype=torch.long)
+        pos_t = torch.tensor(positions, device=device, dtype=torch.int32)
+        slot_s = torch.tensor(slot_suffix, device=device, dtype=torch.int32)
+        slot_f = torch.tensor(slot_full, device=device, dtype=torch.int32)
+        rows_t = torch.tensor(suffix_rows, device=device, dtype=torch.long)
+        tiles = make_prefill_tiles_cached(full_lens, cached_lens, device)
+
+        hidden = self.model.forward_prefill_cached(
+            ids_t, pos_t, tiles, self.kv, slot_s, slot_f, rows_t, full_row)
+        last_hidden = hidden[torch.tensor(last_rows, device=device, dtype=torch.long)]
+        self._sample_and_append(batch, last_hidden)
         return len(input_ids)
 
     def _preempt(self, seq: Sequence) -> None:
@@ -261,8 +352,10 @@ class LLMEngine:
         if seq in self.running:
             self.running.remove(seq)
         if seq.pages:
-            self.kv.free(seq.pages)
+            self._release_pages(seq.pages)
             seq.pages = []
+        seq.cached_len = 0
+        seq.n_cached_pages = 0
         seq.state = SeqState.WAITING
         self.waiting.insert(0, seq)
         logger_warning = getattr(self, "_preempt_count", 0) + 1
@@ -273,7 +366,7 @@ class LLMEngine:
         even after preempting younger sequences."""
         while KVCache.pages_needed(pos + 1) > len(seq.pages):
             try:
-                seq.pages.extend(self.kv.alloc(1))
+                seq.pages.extend(self._alloc_pages(1))
             except MemoryError:
                 victim = None
                 for cand in reversed(self.running):
@@ -481,7 +574,13 @@ class LLMEngine:
         seq.state = SeqState.FINISHED
         seq.finish_reason = reason
         if seq.pages:
-            self.kv.free(seq.pages)
+            if self.prefix_cache is not None and reason in ("stop", "length"):
+                # make this sequence's full pages (prompt + generated)
+                # available to future requests — multi-turn rollouts extend
+                # exactly this token prefix
+                self.prefix_cache.publish(seq.prompt_ids + seq.output_ids,
+                                          seq.pages, seq.n_cached_pages)
+            self._release_pages(seq.pages)
             seq.pages = []
         self.finished[seq.request_id] = seq
 

@@ -185,6 +185,26 @@ class QwenLayer(nn.Module):
         return self._finish(hidden, attn)
 
     @torch.no_grad()
+    def forward_prefill_cached(self, hidden, positions, cos_t, sin_t, tiles, kv_cache,
+                               slot_suffix, slot_full, suffix_rows, T_full, layer_idx):
+        """Prefix-cached prefill: `hidden`/`positions` cover only the SUFFIX
+        tokens (absolute positions); cached prefix K/V already live in the
+        pages. Suffix K/V are written, full-length K/V are gathered back
+        from pages, and the flash kernel runs with q-tiles restricted to the
+        suffix (tile_row0 starts past the cached region)."""
+        cfg = self.cfg
+        q, k, v = self._qkv(hidden, positions, cos_t, sin_t)
+        k_pages, v_pages = kv_cache[layer_idx]
+        ops.reshape_and_cache(k, v, k_pages, v_pages, slot_suffix)
+        k_full, v_full = ops.gather_cache(k_pages, v_pages, slot_full)
+        q_full = q.new_zeros(T_full, cfg.num_heads, cfg.head_dim)
+        q_full.index_copy_(0, suffix_rows, q)
+        attn_full = ops.flash_prefill(q_full, k_full, v_full, tiles[0], tiles[1], tiles[2],
+                                      1.0 / math.sqrt(cfg.head_dim))
+        attn = attn_full.index_select(0, suffix_rows)
+        return self._finish(hidden, attn)
+
+    @torch.no_grad()
     def forward_decode_fused(self, h, delta, positions, cos_t, sin_t, kv_cache,
                              slot_mapping, block_tables, seq_lens, layer_idx):
         """Decode layer with fused residual chaining: returns the next delta.
@@ -317,6 +337,19 @@ class QwenModel(nn.Module):
         return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
 
     @torch.no_grad()
+    def forward_prefill_cached(self, input_ids, positions, tiles, kv_cache,
+                               slot_suffix, slot_full, suffix_rows, T_full):
+        """Suffix-only prefill against prefix-cached pages (engine prefix
+        caching; see QwenLayer.forward_prefill_cached)."""
+        self._stamp_lora()
+        hidden = self.embed_tokens[input_ids]
+        for i, layer in enumerate(self.layers):
+            hidden = layer.forward_prefill_cached(hidden, positions, self.cos_t, self.sin_t,
+                                                  tiles, kv_cache, slot_suffix, slot_full,
+                                                  suffix_rows, T_full, i)
+        return ops.rmsnorm(hidden, self.norm, self.cfg.rms_eps)
+
+    @torch.no_grad()
     def forward_decode(self, input_ids, positions, kv_cache, slot_mapping, block_tables, seq_lens):
         self._stamp_lora()
         h = self.embed_tokens[input_ids].contiguous()
@@ -372,6 +405,23 @@ def make_prefill_tiles(seqlens: list[int], device, tile: int = 64):
     start = 0
     for n in seqlens:
         for r0 in range(0, n, tile):
+            tile_seq_start.append(start)
+            tile_row0.append(start + r0)
+            tile_seq_len.append(n)
+        start += n
+    mk = lambda x: torch.tensor(x, device=device, dtype=torch.int32)
+    return mk(tile_seq_start), mk(tile_row0), mk(tile_seq_len)
+
+
+def make_prefill_tiles_cached(full_lens: list[int], cached_lens: list[int], device, tile: int = 64):
+    """Q-tile table covering only each sequence's SUFFIX (rows past the
+    prefix-cached region), in FULL-length packed coordinates. The flash
+    kernel's q_local0 = row0 - seq_start handles arbitrary suffix starts;
+    its epilogue skips rows below the tile start."""
+    tile_seq_start, tile_row0, tile_seq_len = [], [], []
+    start = 0
+    for n, c in zip(full_lens, cached_lens):
+        for r0 in range(c, n, tile):
             tile_seq_start.append(start)
             tile_row0.append(start + r0)
             tile_seq_len.append(n)

@@ -285,6 +285,11 @@ def reshape_and_cache(k, v, k_pages, v_pages, slot_mapping):
     return require_ext().reshape_and_cache(k, v, k_pages, v_pages, slot_mapping)
 
 
+def gather_cache(k_pages, v_pages, slot_mapping):
+    """Inverse of reshape_and_cache: contiguous [T, Hk, D] K/V from pages."""
+    return require_ext().gather_cache(k_pages, v_pages, slot_mapping)
+
+
 def qkv_rope_cache(qkv, bias, k_pages, v_pages, cos_tab, sin_tab, positions, slot_mapping, Hq, Hk):
     """Fused bias + rope + paged KV write + contiguous-q extract (rollout path)."""
     return require_ext().qkv_rope_cache(qkv, bias, k_pages, v_pages, cos_tab, sin_tab,

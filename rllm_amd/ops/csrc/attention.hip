@@ -456,6 +456,33 @@ __global__ void reshape_and_cache_kernel(
   }
 }
 
+// gather_cache: inverse of reshape_and_cache — collect K/V rows for a run of
+// slots into contiguous [T, Hk, 128] buffers (prefix-cached prefill: the
+// suffix attends against full-length K/V rebuilt from pages).
+__global__ void gather_cache_kernel(
+    const __bf16* __restrict__ Kp,
+    const __bf16* __restrict__ Vp,
+    __bf16* __restrict__ K,
+    __bf16* __restrict__ V,
+    const int32_t* __restrict__ slot_mapping, // [T]
+    int64_t T, int Hk) {
+  const int64_t total = T * Hk * (HEAD_DIM / 8);
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int c8 = (int)(idx % (HEAD_DIM / 8)) * 8;
+    const int64_t th = idx / (HEAD_DIM / 8);
+    const int h = (int)(th % Hk);
+    const int64_t t = th / Hk;
+    const int32_t slot = slot_mapping[t];
+    if (slot < 0) continue;
+    const int page = slot >> 4, off = slot & 15;
+    const int64_t dst = ((int64_t)t * Hk + h) * HEAD_DIM + c8;
+    const int64_t src = (((int64_t)page * Hk + h) * PAGE_SIZE + off) * HEAD_DIM + c8;
+    *reinterpret_cast<bf16x8*>(K + dst) = *reinterpret_cast<const bf16x8*>(Kp + src);
+    *reinterpret_cast<bf16x8*>(V + dst) = *reinterpret_cast<const bf16x8*>(Vp + src);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Host wrappers
 // ---------------------------------------------------------------------------
@@ -616,6 +643,22 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor
     HIP_CHECK_KERNEL();
   }
   return o;
+}
+
+std::vector<torch::Tensor> gather_cache(torch::Tensor k_pages, torch::Tensor v_pages,
+                                        torch::Tensor slot_mapping) {
+  TORCH_CHECK(slot_mapping.dtype() == torch::kInt32 && slot_mapping.is_cuda());
+  const int64_t T = slot_mapping.size(0);
+  const int Hk = (int)k_pages.size(1);
+  auto k = torch::empty({T, Hk, HEAD_DIM}, k_pages.options());
+  auto v = torch::empty({T, Hk, HEAD_DIM}, v_pages.options());
+  const int64_t total = T * Hk * (HEAD_DIM / 8);
+  hipLaunchKernelGGL(gather_cache_kernel, dim3(grid_for(total, 256)), dim3(256), 0, at_stream(),
+                     (const __bf16*)k_pages.data_ptr(), (const __bf16*)v_pages.data_ptr(),
+                     (__bf16*)k.data_ptr(), (__bf16*)v.data_ptr(),
+                     slot_mapping.data_ptr<int32_t>(), T, Hk);
+  HIP_CHECK_KERNEL();
+  return {k, v};
 }
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,

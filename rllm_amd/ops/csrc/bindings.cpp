@@ -45,6 +45,8 @@ std::vector<torch::Tensor> flash_train_bwd(torch::Tensor q, torch::Tensor k, tor
 torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor v_pages,
                            torch::Tensor block_tables, torch::Tensor seq_lens, double scale,
                            int64_t n_splits);
+std::vector<torch::Tensor> gather_cache(torch::Tensor k_pages, torch::Tensor v_pages,
+                                        torch::Tensor slot_mapping);
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_pages,
                        torch::Tensor v_pages, torch::Tensor slot_mapping);
 torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bias,
@@ -82,6 +84,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_train_fwd", &flash_train_fwd, "Training flash attention forward (O + LSE)");
   m.def("flash_train_bwd", &flash_train_bwd, "Training flash attention backward (dQ,dK,dV)");
   m.def("paged_decode", &paged_decode, "Paged decode attention (flash-decoding splits)");
+  m.def("gather_cache", &gather_cache, "Gather K/V rows from pages into contiguous buffers");
   m.def("reshape_and_cache", &reshape_and_cache, "Scatter K/V into KV pages");
   m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");
   m.def("add_rmsnorm_", &add_rmsnorm_, "Fused residual add (in-place) + RMSNorm");
