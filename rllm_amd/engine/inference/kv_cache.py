@@ -72,13 +72,11 @@ class KVCache:
 
 
 def _page_hash(parent: int, tokens: tuple) -> int:
+    import numpy as np
     import xxhash
 
-    h = xxhash.xxh3_64()
-    h.update(parent.to_bytes(8, "little"))
-    for t in tokens:
-        h.update(int(t).to_bytes(4, "little", signed=True))
-    return h.intdigest()
+    return xxhash.xxh3_64_intdigest(
+        parent.to_bytes(8, "little") + np.asarray(tokens, dtype=np.int64).tobytes())
 
 
 class PrefixCache:
@@ -105,11 +103,20 @@ class PrefixCache:
     # -- hashing ------------------------------------------------------------
     @staticmethod
     def page_hashes(tokens: list[int]) -> list[int]:
-        """Hash per FULL page of the token prefix."""
+        """Chained hash per FULL page of the token prefix (one xxh3 C call
+        per page over the page's int64 bytes + parent digest)."""
+        import numpy as np
+        import xxhash
+
+        n_full = len(tokens) // PAGE_SIZE
+        if n_full == 0:
+            return []
+        arr = np.asarray(tokens[: n_full * PAGE_SIZE], dtype=np.int64)
         out = []
         h = 0
-        for i in range(0, len(tokens) - PAGE_SIZE + 1, PAGE_SIZE):
-            h = _page_hash(h, tuple(tokens[i : i + PAGE_SIZE]))
+        for i in range(n_full):
+            h = xxhash.xxh3_64_intdigest(
+                h.to_bytes(8, "little") + arr[i * PAGE_SIZE : (i + 1) * PAGE_SIZE].tobytes())
             out.append(h)
         return out
 

@@ -134,7 +134,9 @@ def test_engine_eos_and_abort():
     assert a.finish_reason in ("stop", "length")
     if a.finish_reason == "stop":
         assert a.token_ids[-1] == 0
-    # KV fully reclaimed
+    # KV fully reclaimed (idle prefix-cache pages are evictable, not leaked)
+    if engine.prefix_cache is not None:
+        engine.prefix_cache.evict(1 << 30)
     assert engine.kv.num_free_pages == engine.kv.num_pages - 1
 
 
@@ -196,6 +198,8 @@ def test_kv_preemption_and_resume():
     outs = engine.generate(prompts, SamplingParams(temperature=1.0, max_tokens=64))
     assert all(len(o.token_ids) == 64 for o in outs)
     assert all(len(o.logprobs) == 64 for o in outs)
+    if engine.prefix_cache is not None:
+        engine.prefix_cache.evict(1 << 30)
     assert engine.kv.num_free_pages == engine.kv.num_pages - 1
 
 
