@@ -122,6 +122,16 @@ class QwenLayer(nn.Module):
     # by QwenModel from its enabled flag.
     lora_scale: float = 1.0
     _lora_on: bool = False
+    # Rollout tensor parallelism (parallel/tp.py enable_tp): o_proj and
+    # down_proj are row-parallel — their partial outputs are all-reduced.
+    tp_group = None
+
+    def _tp_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.tp_group is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(t, group=self.tp_group)
+        return t
 
     def _lin(self, x: torch.Tensor, name: str, b: torch.Tensor | None = None) -> torch.Tensor:
         y = _linear(x, getattr(self, name), b)
@@ -146,9 +156,9 @@ class QwenLayer(nn.Module):
     def _finish(self, hidden: torch.Tensor, attn_out: torch.Tensor) -> torch.Tensor:
         cfg = self.cfg
         T = hidden.shape[0]
-        hidden = hidden + self._lin(attn_out.reshape(T, cfg.q_size), "o_proj")
+        hidden = hidden + self._tp_reduce(self._lin(attn_out.reshape(T, cfg.q_size), "o_proj"))
         x = ops.rmsnorm(hidden, self.post_attention_layernorm, cfg.rms_eps)
-        mlp = self._lin(ops.swiglu(self._lin(x, "gate_up_proj")), "down_proj")
+        mlp = self._tp_reduce(self._lin(ops.swiglu(self._lin(x, "gate_up_proj")), "down_proj"))
         return hidden + mlp
 
     # -- training path -------------------------------------------------------
@@ -220,9 +230,10 @@ class QwenLayer(nn.Module):
                                positions, slot_mapping, cfg.num_heads, cfg.num_kv_heads)
         attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
                                 1.0 / math.sqrt(cfg.head_dim))
-        attn_delta = self._lin_decode(attn.reshape(T, cfg.q_size), "o_proj")
+        attn_delta = self._tp_reduce(self._lin_decode(attn.reshape(T, cfg.q_size), "o_proj"))
         x = ops.add_rmsnorm_(h, attn_delta, self.post_attention_layernorm, cfg.rms_eps)
-        mlp_delta = self._lin_decode(ops.swiglu(self._lin_decode(x, "gate_up_proj")), "down_proj")
+        mlp_delta = self._tp_reduce(
+            self._lin_decode(ops.swiglu(self._lin_decode(x, "gate_up_proj")), "down_proj"))
         return mlp_delta
 
     def _lin_decode(self, x: torch.Tensor, name: str) -> torch.Tensor:

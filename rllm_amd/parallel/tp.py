@@ -122,3 +122,38 @@ def tp_reference_forward(full_weight_matmul, shards: list[torch.Tensor], x: torc
             outs = o if outs is None else outs + o
         return outs
     raise ValueError(mode)
+
+
+# ---------------------------------------------------------------------------
+# Runtime wiring (rollout TP): sharded model construction + per-layer
+# all-reduce hooks. The forward hooks live in models/qwen.py (`tp_group` on
+# each layer: o_proj and down_proj partials are summed over xGMI).
+# ---------------------------------------------------------------------------
+
+
+def build_tp_model(cfg: ModelConfig, tp_rank: int, tp: int, *, device: str = "cuda",
+                   seed: int = 0, tp_group=None, full_state_dict: dict | None = None):
+    """One TP rank's QwenModel: identical full-model init on every rank
+    (same seed) sliced to this rank's shard, so all ranks agree on the
+    replicated tensors bit-for-bit (embeddings/lm_head/norms — required
+    for lockstep sampling)."""
+    from rllm_amd.models.qwen import QwenModel
+
+    if full_state_dict is None:
+        full = QwenModel(cfg, device="cpu").init_random(seed=seed)
+        full_state_dict = full.state_dict()
+    shard_cfg = shard_model_config(cfg, tp)
+    sd = shard_state_dict(full_state_dict, cfg, tp_rank, tp)
+    model = QwenModel(shard_cfg, device=device)
+    model.load_state_dict({k: v.to(device) for k, v in sd.items()})
+    enable_tp(model, tp_group)
+    return model
+
+
+def enable_tp(model, tp_group) -> None:
+    """Arm the per-layer all-reduce hooks (rollout paths only: the hooks
+    fire in _finish / forward_decode_fused; the training path must not run
+    under TP — gradients would need their own collective plan)."""
+    model.tp_group = tp_group
+    for layer in model.layers:
+        layer.tp_group = tp_group

@@ -107,3 +107,105 @@ def test_shard_state_dict_shapes():
     assert out["layers.0.gate_up_proj"].shape == (2 * scfg.intermediate_size, cfg.hidden_size)
     assert out["layers.0.down_proj"].shape == (cfg.hidden_size, scfg.intermediate_size)
     assert out["embed_tokens"].shape == sd["embed_tokens"].shape  # replicated
+
+
+def test_build_tp_model_cpu_shapes_and_replication():
+    from rllm_amd.models.config import ModelConfig
+    from rllm_amd.parallel.tp import build_tp_model, shard_qkv
+
+    cfg = ModelConfig(name="tpt", hidden_size=256, intermediate_size=512,
+                      num_layers=2, num_heads=4, num_kv_heads=2, head_dim=64,
+                      vocab_size=128, tie_word_embeddings=False)
+    import torch
+
+    from rllm_amd.models.qwen import QwenModel
+
+    full = QwenModel(cfg, device="cpu").init_random(seed=6)
+    fsd = full.state_dict()
+    m0 = build_tp_model(cfg, 0, 2, device="cpu", full_state_dict=fsd)
+    m1 = build_tp_model(cfg, 1, 2, device="cpu", full_state_dict=fsd)
+    # sharded dims
+    assert m0.layers[0].qkv_proj.shape == ((4 // 2 + 2) * 64, 256)  # 2 q + 1+1 kv heads... see below
+    assert m0.layers[0].o_proj.shape == (256, 2 * 64)
+    assert m0.layers[0].gate_up_proj.shape == (512, 256)  # 2 * I/tp
+    assert m0.layers[0].down_proj.shape == (256, 256)
+    # replicated tensors identical across ranks
+    assert torch.equal(m0.embed_tokens, m1.embed_tokens)
+    assert torch.equal(m0.lm_head, m1.lm_head)
+    # shard content matches the slicing plan
+    expect0 = shard_qkv(fsd["layers.0.qkv_proj"], cfg, 0, 2)
+    assert torch.equal(m0.layers[0].qkv_proj, expect0)
+    # column shards partition the full weight
+    cat_o = torch.cat([m0.layers[0].o_proj, m1.layers[0].o_proj], dim=1)
+    assert torch.equal(cat_o, fsd["layers.0.o_proj"])
+
+
+def _worker_tp_reduce(rank, world, q):
+    try:
+        pdist = _setup_dist(rank, world)
+        import torch
+
+        from rllm_amd.models.config import ModelConfig
+        from rllm_amd.models.qwen import QwenModel
+        from rllm_amd.parallel.tp import build_tp_model
+
+        cfg = ModelConfig(name="tpr", hidden_size=128, intermediate_size=256,
+                          num_layers=1, num_heads=4, num_kv_heads=2, head_dim=32,
+                          vocab_size=64, tie_word_embeddings=False)
+        full = QwenModel(cfg, device="cpu").init_random(seed=9)
+        import torch.distributed as dist
+
+        model = build_tp_model(cfg, rank, world, device="cpu",
+                               full_state_dict=full.state_dict(),
+                               tp_group=dist.group.WORLD)
+        layer = model.layers[0]
+        assert layer.tp_group is not None
+        # row-parallel o_proj: partial per rank, all_reduce == full product
+        torch.manual_seed(0)
+        x_full = torch.randn(3, cfg.q_size).to(torch.bfloat16)
+        qs = cfg.q_size // world
+        partial = (x_full[:, rank * qs : (rank + 1) * qs].float()
+                   @ layer.o_proj.float().t())
+        layer._tp_reduce(partial)
+        expect = x_full.float() @ full.layers[0].o_proj.float().t()
+        assert torch.allclose(partial, expect, atol=1e-2), (partial - expect).abs().max()
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, f"FAIL: {traceback.format_exc()[-400:]}"))
+
+
+def test_tp_reduce_world2():
+    import os
+
+    import torch.multiprocessing as mp
+
+    os.environ["TEST_DIST_PORT"] = "29625"
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_tp_reduce, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(), q.get()]
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
+
+
+def _setup_dist(rank, world):
+    import os
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = os.environ.get("TEST_DIST_PORT", "29625")
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    from rllm_amd.parallel import dist as pdist
+
+    pdist.init_from_env(backend="gloo")
+    return pdist
