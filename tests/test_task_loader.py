@@ -61,3 +61,24 @@ def test_missing_dataset_raises(tmp_path):
 
     with pytest.raises(FileNotFoundError):
         load_tasks(tmp_path / "nope")
+
+
+def test_build_swe_dataset(tmp_path):
+    import json
+
+    from rllm_amd.data.builders import build_swe_dataset
+    from rllm_amd.data.dataset import DatasetRegistry
+
+    rows = [{"instance_id": "astropy__astropy-123", "problem_statement": "fix the bug",
+             "repo": "astropy/astropy", "base_commit": "abc123",
+             "patch": "diff --git ...", "FAIL_TO_PASS": ["test_a"], "PASS_TO_PASS": ["test_b"]}]
+    p = tmp_path / "swe.jsonl"
+    p.write_text("\n".join(json.dumps(r) for r in rows))
+    reg = DatasetRegistry(root=str(tmp_path / "reg"))
+    ds = build_swe_dataset(p, reg, name="swe-mini")
+    assert len(ds) == 1
+    row = ds[0]
+    assert row["id"] == "astropy__astropy-123"
+    assert row["question"] == "fix the bug"
+    assert row["fail_to_pass"] == ["test_a"]
+    assert reg.load_dataset("swe-mini") is not None
