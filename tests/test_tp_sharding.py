@@ -209,3 +209,101 @@ def _setup_dist(rank, world):
 
     pdist.init_from_env(backend="gloo")
     return pdist
+
+
+class _FakeEngine:
+    """Deterministic stand-in for LLMEngine: echoes prompt-derived tokens
+    over a fixed number of steps so two lockstep ranks can be compared."""
+
+    def __init__(self):
+        self.seqs = {}
+        self.log = []
+        self.weight_version = 0
+        self.finished = {}
+
+    def add_request(self, rid, prompt, params):
+        self.log.append(("add", rid, tuple(prompt)))
+        self.seqs[rid] = {"prompt": list(prompt), "out": [], "n": params}
+
+    def abort(self, rid):
+        self.log.append(("abort", rid))
+        self.seqs.pop(rid, None)
+
+    def has_unfinished(self):
+        return bool(self.seqs)
+
+    def step(self):
+        self.log.append(("step",))
+        done = []
+        for rid, s in self.seqs.items():
+            s["out"].append(sum(s["prompt"]) % 97 + len(s["out"]))
+            if len(s["out"]) >= s["n"]:
+                done.append(rid)
+        for rid in done:
+            self.finished[rid] = self.seqs.pop(rid)
+        return len(self.seqs) + len(done)
+
+    def pop_finished(self):
+        import types
+
+        outs = []
+        for rid, s in self.finished.items():
+            outs.append(types.SimpleNamespace(request_id=rid, token_ids=s["out"]))
+        self.finished = {}
+        return outs
+
+
+def _worker_tp_lockstep(rank, world, q):
+    try:
+        pdist = _setup_dist(rank, world)
+        import torch.distributed as dist
+
+        from rllm_amd.parallel.tp_runner import TPLockstepEngine
+
+        eng = _FakeEngine()
+        wrap = TPLockstepEngine(eng, tp_group=dist.group.WORLD)
+        dist = __import__("torch.distributed", fromlist=["x"])
+        if wrap.is_driver:
+            wrap.add_request("a", [3, 4], 3)
+            wrap.add_request("b", [10], 2)
+            wrap.step()
+            wrap.set_weight_version(7)
+            wrap.abort("b")
+            while wrap.has_unfinished():
+                wrap.step()
+            outs = {o.request_id: o.token_ids for o in wrap.pop_finished()}
+            assert outs == {"a": [7, 8, 9]}, outs
+            wrap.stop()
+        else:
+            wrap.serve()
+            assert eng.weight_version == 7
+        # both ranks saw the identical command/step stream
+        gathered = [None] * world
+        dist.all_gather_object(gathered, eng.log)
+        assert gathered[0] == gathered[1], (gathered[0], gathered[1])
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, f"FAIL: {traceback.format_exc()[-400:]}"))
+
+
+def test_tp_lockstep_world2():
+    import os
+
+    import torch.multiprocessing as mp
+
+    os.environ["TEST_DIST_PORT"] = "29627"
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_tp_lockstep, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(), q.get()]
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
