@@ -60,6 +60,7 @@ class NativeBackend(BackendProtocol):
         checkpoint_path: str | None = None,
         seed: int = 0,
         lora=None,  # models.lora.LoRAConfig | True for defaults | None = full finetune
+        gateway_config=None,  # GatewayConfig; e.g. cumulative_token_mode=True for multi-turn
     ):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
@@ -78,6 +79,7 @@ class NativeBackend(BackendProtocol):
         self.n_parallel_tasks = n_parallel_tasks
         self.checkpoint_path = checkpoint_path
         self.lora_config = lora
+        self.gateway_config = gateway_config
 
         self.model: QwenModel | None = None
         self.ref_model: QwenModel | None = None
@@ -136,7 +138,10 @@ class NativeBackend(BackendProtocol):
                                            train_sampling_params=self.rollout_sampling_params)
 
         handler = make_native_local_handler(self.driver, self.parser, self.cfg.name)
-        self.gateway = GatewayManager(GatewayConfig(), local_handler=handler)
+        gw_cfg = self.gateway_config or GatewayConfig()
+        # cumulative token mode needs the parser for prompt-id rebuilding
+        self.gateway = GatewayManager(gw_cfg, local_handler=handler,
+                                      parser=self.parser if gw_cfg.cumulative_token_mode else None)
         self.gateway.start()
         self.flow_engine = AgentFlowEngine(
             self.agent_flow, self.gateway, model_name=self.cfg.name,
