@@ -311,6 +311,13 @@ import os as _os
 _USE_SKINNY = _os.environ.get("RLLM_SKINNY_GEMM", "0") == "1"
 
 
+def skinny_gemm_v3(x, w, bias=None):
+    """v3 hand-written skinny GEMM (64x64 wave tiles, double-buffered LDS,
+    partials split-K). Experimental: enable per-shape only where it beats
+    the tuned hipBLASLt path (see profiles/)."""
+    return require_ext().skinny_gemm_v3(x, w, bias)
+
+
 def linear_decode(x, w, bias=None):
     """Decode-path linear. Default: hipBLASLt with per-shape TUNED algo
     selection (hbl_tuned.hip) when `pretune_decode_shapes` has run for this

@@ -57,6 +57,7 @@ torch::Tensor qkv_rope_cache(torch::Tensor qkv, c10::optional<torch::Tensor> bia
 
 // skinny_gemm.hip
 torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, c10::optional<torch::Tensor> bias);
+torch::Tensor skinny_gemm_v3(torch::Tensor a, torch::Tensor w, c10::optional<torch::Tensor> bias);
 
 // hbl_tuned.hip
 std::vector<double> hbl_tune(torch::Tensor x, torch::Tensor w, int64_t iters);
@@ -89,6 +90,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_cache", &qkv_rope_cache, "Fused bias+rope+cache-write+q-extract");
   m.def("add_rmsnorm_", &add_rmsnorm_, "Fused residual add (in-place) + RMSNorm");
   m.def("skinny_gemm", &skinny_gemm, "Weight-streaming skinny-M GEMM (decode path)");
+  m.def("skinny_gemm_v3", &skinny_gemm_v3, "Skinny GEMM v3: 64x64 wave tiles, double-buffered LDS, partials split-K");
   m.def("hbl_tune", &hbl_tune, "Sweep hipblaslt heuristic algos for a decode shape; cache winner");
   m.def("hbl_has", &hbl_has, "Is this (M,N,K) tuned?");
   m.def("hbl_mm", &hbl_mm, "Tuned hipblaslt GEMM (graph-capture safe)");
