@@ -107,7 +107,7 @@ class LLMEngine:
                  max_num_seqs: int = 1024, max_num_batched_tokens: int = 8192,
                  kv_budget_bytes: int | None = None, eos_token_id: int | None = None,
                  seed: int = 0, use_hip_graph: bool = True, max_model_len: int = 8192,
-                 enable_prefix_caching: bool = True):
+                 enable_prefix_caching: bool = True, fp8_decode: bool | None = None):
         self.model = model
         self.cfg = model.cfg
         self.device = next(model.parameters()).device
@@ -128,6 +128,15 @@ class LLMEngine:
         # automatic prefix caching: cross-request page reuse keyed on token
         # content (multi-turn/cumulative rollouts share prompt prefixes)
         self.prefix_cache = PrefixCache(self.kv) if enable_prefix_caching else None
+        # fp8 rollout (opt-in, RLLM_FP8_ROLLOUT=1): e4m3 decode GEMMs;
+        # weights re-quantized on every weight-version bump
+        if fp8_decode is None:
+            import os as _os
+
+            fp8_decode = _os.environ.get("RLLM_FP8_ROLLOUT", "0") == "1"
+        self.fp8_decode = fp8_decode
+        if fp8_decode:
+            self.model.enable_fp8_decode()
 
         self.waiting: list[Sequence] = []
         self.running: list[Sequence] = []
@@ -152,9 +161,12 @@ class LLMEngine:
 
     @weight_version.setter
     def weight_version(self, v: int) -> None:
-        if v != self._weight_version and self.prefix_cache is not None:
-            # cached KV was computed under the old weights — stale
-            self.prefix_cache.clear()
+        if v != self._weight_version:
+            if self.prefix_cache is not None:
+                # cached KV was computed under the old weights — stale
+                self.prefix_cache.clear()
+            if self.fp8_decode:
+                self.model.enable_fp8_decode()  # re-quantize fresh weights
         self._weight_version = v
 
     def _alloc_pages(self, n: int) -> list[int]:
@@ -465,6 +477,11 @@ class LLMEngine:
         # capture — the default heuristic has a ~20 us/call floor at skinny M
         ops.pretune_decode_shapes(ops.decode_gemm_shapes(self.cfg, [Bp]),
                                   device=self.device)
+        if self.fp8_decode:
+            # projections only — logits/sampling stay bf16
+            ops.pretune_fp8_decode_shapes(
+                [s for s in ops.decode_gemm_shapes(self.cfg, [Bp])
+                 if s[1] != self.cfg.vocab_size], device=self.device)
         # warm up eager once (cuBLAS-style workspace init), then capture
         torch.cuda.synchronize()
         self._decode_compute(Bp, temp)

@@ -237,10 +237,17 @@ class QwenLayer(nn.Module):
         return mlp_delta
 
     def _lin_decode(self, x: torch.Tensor, name: str) -> torch.Tensor:
-        """Decode-path linear: hipBLASLt skinny GEMM + optional unmerged-LoRA
-        term (rollouts normally run MERGED — merge_lora_ — so this branch is
-        off; kept for logprob-exactness checks against the train path)."""
-        y = ops.linear_decode(x, getattr(self, name))
+        """Decode-path linear: hipBLASLt skinny GEMM (tuned algo; fp8 e4m3
+        weights+activations when enable_fp8_decode is armed — rollout-only,
+        the bf16 update corrects the distribution shift via TIS) + optional
+        unmerged-LoRA term (rollouts normally run MERGED — merge_lora_ — so
+        that branch is off; kept for logprob-exactness checks)."""
+        fp8 = getattr(self, "_fp8", None)
+        if fp8 is not None and name in fp8:
+            w8, sw = fp8[name]
+            y = ops.fp8_linear(x, w8, sw, w_bf16=getattr(self, name))
+        else:
+            y = ops.linear_decode(x, getattr(self, name))
         if self._lora_on and f"{name}_A" in self.lora:
             from rllm_amd.models import lora as _lora
 
@@ -269,6 +276,24 @@ class QwenModel(nn.Module):
     @property
     def lm_weight(self) -> torch.Tensor:
         return self.embed_tokens if self.lm_head is None else self.lm_head
+
+    def enable_fp8_decode(self) -> None:
+        """(Re)quantize the four decode projections to OCP e4m3 with
+        per-tensor scales. Call after every weight update (the engine's
+        weight_version setter does) — the fp8 copies alias nothing."""
+        with torch.no_grad():
+            for layer in self.layers:
+                q = {}
+                for name in ("qkv_proj", "o_proj", "gate_up_proj", "down_proj"):
+                    q[name] = ops.fp8_quant(getattr(layer, name).detach())
+                layer._fp8 = q
+        self.fp8_decode = True
+
+    def disable_fp8_decode(self) -> None:
+        for layer in self.layers:
+            if hasattr(layer, "_fp8"):
+                del layer._fp8
+        self.fp8_decode = False
 
     def _stamp_lora(self) -> None:
         """Propagate the model-level adapter state to the layers once per

@@ -311,6 +311,60 @@ import os as _os
 _USE_SKINNY = _os.environ.get("RLLM_SKINNY_GEMM", "0") == "1"
 
 
+def fp8_quant(t: torch.Tensor):
+    """Per-tensor OCP e4m3 quantization -> (t8, scale [1] fp32 on device);
+    original ≈ t8.float() * scale. Graph-capturable (amax/div/cast only)."""
+    scale = (t.float().abs().amax() / 448.0).clamp(min=1e-8).reshape(1)
+    t8 = (t.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return t8.contiguous(), scale
+
+
+def fp8_linear(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
+               w_bf16: torch.Tensor | None = None):
+    """y = x @ dequant(w8)^T with dynamic per-call activation quantization.
+    Requires the (M,N,K) shape fp8-tuned (pretune_fp8_decode_shapes);
+    untuned shapes fall back to the bf16 weight (w_bf16) when provided."""
+    C = require_ext()
+    M, K = x.shape[0], x.shape[-1]
+    N = w8.shape[0]
+    if not C.hbl_fp8_has(M, N, K):
+        if w_bf16 is not None:
+            return linear_decode(x, w_bf16)
+        raise RuntimeError(f"fp8 shape {(M, N, K)} not tuned")
+    x8, sx = fp8_quant(x)
+    return C.hbl_fp8_mm(x8, w8, sx, sw)
+
+
+_FP8_TUNED: set = set()
+
+
+def pretune_fp8_decode_shapes(shapes, device="cuda", iters: int = 20, verbose: bool = True):
+    """fp8 analogue of pretune_decode_shapes. Returns {shape: best_us};
+    shapes that produce no valid fp8 algo are skipped (bf16 fallback)."""
+    if _TUNE_DISABLED:
+        return {}
+    C = require_ext()
+    out = {}
+    for (M, N, K) in shapes:
+        key = (int(M), int(N), int(K))
+        if key in _FP8_TUNED:
+            continue
+        x8, sx = fp8_quant(torch.randn(M, K, device=device))
+        w8, sw = fp8_quant(torch.randn(N, K, device=device))
+        try:
+            best_us, n_cand, n_valid, idx = C.hbl_fp8_tune(x8, w8, sx, sw, iters)
+        except RuntimeError as e:
+            if verbose:
+                print(f"[ops.pretune-fp8] {key}: no valid algo ({e}); bf16 fallback")
+            continue
+        _FP8_TUNED.add(key)
+        out[key] = best_us
+        if verbose:
+            print(f"[ops.pretune-fp8] {key}: {best_us:.2f} us "
+                  f"(algo {int(idx)}, {int(n_valid)}/{int(n_cand)} valid)")
+    return out
+
+
 def skinny_gemm_v3(x, w, bias=None):
     """v3 hand-written skinny GEMM (64x64 wave tiles, double-buffered LDS,
     partials split-K). Experimental: enable per-shape only where it beats

@@ -63,6 +63,9 @@ torch::Tensor skinny_gemm_v3(torch::Tensor a, torch::Tensor w, c10::optional<tor
 std::vector<double> hbl_tune(torch::Tensor x, torch::Tensor w, int64_t iters);
 bool hbl_has(int64_t M, int64_t N, int64_t K);
 torch::Tensor hbl_mm(torch::Tensor x, torch::Tensor w);
+std::vector<double> hbl_fp8_tune(torch::Tensor x8, torch::Tensor w8, torch::Tensor sx, torch::Tensor sw, int64_t iters);
+bool hbl_fp8_has(int64_t M, int64_t N, int64_t K);
+torch::Tensor hbl_fp8_mm(torch::Tensor x8, torch::Tensor w8, torch::Tensor sx, torch::Tensor sw);
 
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
@@ -94,6 +97,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hbl_tune", &hbl_tune, "Sweep hipblaslt heuristic algos for a decode shape; cache winner");
   m.def("hbl_has", &hbl_has, "Is this (M,N,K) tuned?");
   m.def("hbl_mm", &hbl_mm, "Tuned hipblaslt GEMM (graph-capture safe)");
+  m.def("hbl_fp8_tune", &hbl_fp8_tune, "Tune fp8(e4m3) hipblaslt GEMM for a shape");
+  m.def("hbl_fp8_has", &hbl_fp8_has, "Is this shape fp8-tuned?");
+  m.def("hbl_fp8_mm", &hbl_fp8_mm, "Tuned fp8 GEMM with device scale pointers");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

@@ -36,8 +36,9 @@ namespace {
 
 struct ShapeKey {
   int M, N, K;
+  int fp8 = 0;
   bool operator<(const ShapeKey& o) const {
-    return std::tie(M, N, K) < std::tie(o.M, o.N, o.K);
+    return std::tie(M, N, K, fp8) < std::tie(o.M, o.N, o.K, o.fp8);
   }
 };
 
@@ -65,7 +66,8 @@ hipblasLtHandle_t handle() {
 
 // Row-major y[M,N] = x[M,K] @ W[N,K]^T as column-major
 // D[N,M] = A^T(W:[K,N],ld K) @ B(x:[K,M],ld K), C=D ld N.
-CachedMatmul make_desc(int M, int N, int K) {
+// fp8: A/B are OCP e4m3 with device scale pointers (set per call), D bf16.
+CachedMatmul make_desc(int M, int N, int K, bool fp8 = false) {
   CachedMatmul cm;
   HBL_CHECK(hipblasLtMatmulDescCreate(&cm.desc, HIPBLAS_COMPUTE_32F, HIP_R_32F));
   int32_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
@@ -73,10 +75,18 @@ CachedMatmul make_desc(int M, int N, int K) {
                                             &opT, sizeof(opT)));
   HBL_CHECK(hipblasLtMatmulDescSetAttribute(cm.desc, HIPBLASLT_MATMUL_DESC_TRANSB,
                                             &opN, sizeof(opN)));
-  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.la, HIP_R_16BF, K, N, K));
-  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.lb, HIP_R_16BF, K, M, K));
+  const hipDataType ab = fp8 ? HIP_R_8F_E4M3 : HIP_R_16BF;
+  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.la, ab, K, N, K));
+  HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.lb, ab, K, M, K));
   HBL_CHECK(hipblasLtMatrixLayoutCreate(&cm.lc, HIP_R_16BF, N, M, N));
   return cm;
+}
+
+void set_scales(const CachedMatmul& cm, const void* sa, const void* sb) {
+  HBL_CHECK(hipblasLtMatmulDescSetAttribute(cm.desc, HIPBLASLT_MATMUL_DESC_A_SCALE_POINTER,
+                                            &sa, sizeof(sa)));
+  HBL_CHECK(hipblasLtMatmulDescSetAttribute(cm.desc, HIPBLASLT_MATMUL_DESC_B_SCALE_POINTER,
+                                            &sb, sizeof(sb)));
 }
 
 void run_matmul(const CachedMatmul& cm, const void* x, const void* w, void* y,
@@ -202,6 +212,116 @@ torch::Tensor hbl_mm(torch::Tensor x, torch::Tensor w) {
   }
   auto y = torch::empty({M, N}, x.options());
   run_matmul(*cm, x.data_ptr(), w.data_ptr(), y.data_ptr(), &cm->algo,
+             at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3) path: x8 [M,K] fp8, w8 [N,K] fp8, per-tensor fp32 device
+// scales sx/sw (the ORIGINAL values are x8*sx etc.); output bf16.
+// ---------------------------------------------------------------------------
+
+std::vector<double> hbl_fp8_tune(torch::Tensor x8, torch::Tensor w8,
+                                 torch::Tensor sx, torch::Tensor sw, int64_t iters) {
+  TORCH_CHECK(x8.is_cuda() && x8.dtype() == torch::kFloat8_e4m3fn && x8.is_contiguous());
+  TORCH_CHECK(w8.is_cuda() && w8.dtype() == torch::kFloat8_e4m3fn && w8.is_contiguous());
+  TORCH_CHECK(sx.is_cuda() && sx.dtype() == torch::kFloat32 && sx.numel() == 1);
+  TORCH_CHECK(sw.is_cuda() && sw.dtype() == torch::kFloat32 && sw.numel() == 1);
+  const int M = (int)x8.size(0), K = (int)x8.size(1), N = (int)w8.size(0);
+  TORCH_CHECK(w8.size(1) == K);
+  hipStream_t stream = at::hip::getCurrentHIPStream().stream();
+
+  std::lock_guard<std::mutex> lk(g_mu);
+  CachedMatmul cm = make_desc(M, N, K, /*fp8=*/true);
+  set_scales(cm, sw.data_ptr(), sx.data_ptr());
+
+  hipblasLtMatmulPreference_t pref;
+  HBL_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  uint64_t ws = kWorkspaceBytes;
+  HBL_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
+  constexpr int kTopN = 48;
+  hipblasLtMatmulHeuristicResult_t results[kTopN];
+  int returned = 0;
+  hipblasStatus_t hs = hipblasLtMatmulAlgoGetHeuristic(
+      handle(), cm.desc, cm.la, cm.lb, cm.lc, cm.lc, pref, kTopN, results, &returned);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  TORCH_CHECK(hs == HIPBLAS_STATUS_SUCCESS && returned > 0,
+              "no fp8 hipblaslt algos for ", M, "x", N, "x", K, " (status ", (int)hs, ")");
+
+  // reference: dequantized fp32 matmul (fp8 rounding already inside x8/w8)
+  auto ref = torch::matmul(x8.to(torch::kFloat32) * sx, (w8.to(torch::kFloat32) * sw).t());
+  const double tol = ref.abs().max().item<double>() * 0.06 + 0.2;
+
+  // cold-weight copy ring
+  const int64_t w_bytes = (int64_t)N * K;
+  const int n_copies = (int)std::min<int64_t>(
+      std::max<int64_t>(1, (256ll << 20) / std::max<int64_t>(1, w_bytes) + 1), 48);
+  std::vector<torch::Tensor> wcopies;
+  wcopies.push_back(w8);
+  for (int i = 1; i < n_copies; ++i) wcopies.push_back(w8.clone());
+
+  auto y = torch::empty({M, N}, x8.options().dtype(torch::kBFloat16));
+  hipEvent_t ev0, ev1;
+  (void)hipEventCreate(&ev0);
+  (void)hipEventCreate(&ev1);
+  double best_us = 1e30;
+  int best_idx = -1, n_valid = 0;
+  float alpha = 1.f, beta = 0.f;
+  for (int i = 0; i < returned; ++i) {
+    if (results[i].state != HIPBLAS_STATUS_SUCCESS) continue;
+    hipblasStatus_t st = hipblasLtMatmul(handle(), cm.desc, &alpha,
+                                         w8.data_ptr(), cm.la, x8.data_ptr(), cm.lb,
+                                         &beta, y.data_ptr(), cm.lc, y.data_ptr(), cm.lc,
+                                         &results[i].algo, g_workspace, kWorkspaceBytes, stream);
+    if (st != HIPBLAS_STATUS_SUCCESS) continue;
+    double err = (y.to(torch::kFloat32) - ref).abs().max().item<double>();
+    if (!(err <= tol)) continue;
+    ++n_valid;
+    for (int r = 0; r < 3; ++r)
+      run_matmul(cm, x8.data_ptr(), wcopies[r % n_copies].data_ptr(), y.data_ptr(),
+                 &results[i].algo, stream);
+    (void)hipEventRecord(ev0, stream);
+    for (int64_t r = 0; r < iters; ++r)
+      run_matmul(cm, x8.data_ptr(), wcopies[r % n_copies].data_ptr(), y.data_ptr(),
+                 &results[i].algo, stream);
+    (void)hipEventRecord(ev1, stream);
+    (void)hipEventSynchronize(ev1);
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, ev0, ev1);
+    double us = ms * 1000.0 / (double)iters;
+    if (us < best_us) { best_us = us; best_idx = i; }
+  }
+  (void)hipEventDestroy(ev0);
+  (void)hipEventDestroy(ev1);
+  TORCH_CHECK(best_idx >= 0, "no valid fp8 algo for ", M, "x", N, "x", K);
+  cm.algo = results[best_idx].algo;
+  cm.valid = true;
+  g_cache[ShapeKey{M, N, K, 1}] = cm;
+  return {best_us, (double)returned, (double)n_valid, (double)best_idx};
+}
+
+bool hbl_fp8_has(int64_t M, int64_t N, int64_t K) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_cache.find(ShapeKey{(int)M, (int)N, (int)K, 1});
+  return it != g_cache.end() && it->second.valid;
+}
+
+torch::Tensor hbl_fp8_mm(torch::Tensor x8, torch::Tensor w8,
+                         torch::Tensor sx, torch::Tensor sw) {
+  const int M = (int)x8.size(0), K = (int)x8.size(1), N = (int)w8.size(0);
+  const CachedMatmul* cm;
+  {
+    std::lock_guard<std::mutex> lk(g_mu);
+    auto it = g_cache.find(ShapeKey{M, N, K, 1});
+    TORCH_CHECK(it != g_cache.end() && it->second.valid,
+                "hbl_fp8_mm: shape ", M, "x", N, "x", K, " not fp8-tuned");
+    cm = &it->second;
+    // scale pointers live on the shared desc: set under the lock per call
+    set_scales(*cm, sw.data_ptr(), sx.data_ptr());
+  }
+  auto y = torch::empty({M, N}, x8.options().dtype(torch::kBFloat16));
+  run_matmul(*cm, x8.data_ptr(), w8.data_ptr(), y.data_ptr(), &cm->algo,
              at::hip::getCurrentHIPStream().stream());
   return y;
 }
