@@ -244,8 +244,9 @@ class QwenLayer(nn.Module):
         that branch is off; kept for logprob-exactness checks)."""
         fp8 = getattr(self, "_fp8", None)
         if fp8 is not None and name in fp8:
-            w8, sw = fp8[name]
-            y = ops.fp8_linear(x, w8, sw, w_bf16=getattr(self, name))
+            w8, sw, scale_buf, amax_buf = fp8[name]
+            y = ops.fp8_linear_delayed(x, w8, sw, scale_buf, amax_buf,
+                                       w_bf16=getattr(self, name))
         else:
             y = ops.linear_decode(x, getattr(self, name))
         if self._lora_on and f"{name}_A" in self.lora:
@@ -283,9 +284,23 @@ class QwenModel(nn.Module):
         weight_version setter does) — the fp8 copies alias nothing."""
         with torch.no_grad():
             for layer in self.layers:
+                old = getattr(layer, "_fp8", None)
                 q = {}
                 for name in ("qkv_proj", "o_proj", "gate_up_proj", "down_proj"):
-                    q[name] = ops.fp8_quant(getattr(layer, name).detach())
+                    w8, sw = ops.fp8_quant(getattr(layer, name).detach())
+                    if old is not None and name in old:
+                        # keep the per-call-site activation-scale state (and
+                        # buffer identity — captured graphs hold its pointer)
+                        _, old_sw, scale_buf, amax_buf = old[name]
+                        old_w8 = old[name][0]
+                        old_w8.copy_(w8)
+                        old_sw.copy_(sw)
+                        q[name] = (old_w8, old_sw, scale_buf, amax_buf)
+                    else:
+                        dev = w8.device
+                        q[name] = (w8, sw,
+                                   torch.ones(1, device=dev, dtype=torch.float32),
+                                   torch.zeros(1, device=dev, dtype=torch.float32))
                 layer._fp8 = q
         self.fp8_decode = True
 

@@ -321,9 +321,9 @@ def fp8_quant(t: torch.Tensor):
 
 def fp8_linear(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
                w_bf16: torch.Tensor | None = None):
-    """y = x @ dequant(w8)^T with dynamic per-call activation quantization.
-    Requires the (M,N,K) shape fp8-tuned (pretune_fp8_decode_shapes);
-    untuned shapes fall back to the bf16 weight (w_bf16) when provided."""
+    """y = x @ dequant(w8)^T with dynamic per-call activation quantization
+    (torch-level; ~6 small kernels — prefer fp8_linear_delayed on hot
+    paths). Untuned shapes fall back to the bf16 weight when provided."""
     C = require_ext()
     M, K = x.shape[0], x.shape[-1]
     N = w8.shape[0]
@@ -333,6 +333,28 @@ def fp8_linear(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
         raise RuntimeError(f"fp8 shape {(M, N, K)} not tuned")
     x8, sx = fp8_quant(x)
     return C.hbl_fp8_mm(x8, w8, sx, sw)
+
+
+def fp8_linear_delayed(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
+                       scale_buf: torch.Tensor, amax_buf: torch.Tensor,
+                       w_bf16: torch.Tensor | None = None):
+    """Hot-path fp8 linear with DELAYED activation scaling: quantize with the
+    previous call's scale while accumulating this call's amax in the same
+    kernel; the scale folds forward after the GEMM (stream-ordered). Three
+    launches total (quant, GEMM, scale-update); graph-capture safe —
+    scale/amax are persistent per-call-site buffers."""
+    C = require_ext()
+    M, K = x.shape[0], x.shape[-1]
+    N = w8.shape[0]
+    if not C.hbl_fp8_has(M, N, K):
+        if w_bf16 is not None:
+            return linear_decode(x, w_bf16)
+        raise RuntimeError(f"fp8 shape {(M, N, K)} not tuned")
+    x8 = torch.empty(x.shape, device=x.device, dtype=torch.float8_e4m3fn)
+    C.fp8_quant_delayed(x.contiguous(), x8, scale_buf, amax_buf)
+    y = C.hbl_fp8_mm(x8, w8, scale_buf, sw)
+    C.fp8_scale_update(scale_buf, amax_buf)
+    return y
 
 
 _FP8_TUNED: set = set()

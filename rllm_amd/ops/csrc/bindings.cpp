@@ -67,6 +67,10 @@ std::vector<double> hbl_fp8_tune(torch::Tensor x8, torch::Tensor w8, torch::Tens
 bool hbl_fp8_has(int64_t M, int64_t N, int64_t K);
 torch::Tensor hbl_fp8_mm(torch::Tensor x8, torch::Tensor w8, torch::Tensor sx, torch::Tensor sw);
 
+// elementwise.hip (fp8)
+void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale, torch::Tensor amax_next);
+void fp8_scale_update(torch::Tensor scale, torch::Tensor amax_next);
+
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
                                           int64_t seed, int64_t step,
@@ -100,6 +104,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hbl_fp8_tune", &hbl_fp8_tune, "Tune fp8(e4m3) hipblaslt GEMM for a shape");
   m.def("hbl_fp8_has", &hbl_fp8_has, "Is this shape fp8-tuned?");
   m.def("hbl_fp8_mm", &hbl_fp8_mm, "Tuned fp8 GEMM with device scale pointers");
+  m.def("fp8_quant_delayed", &fp8_quant_delayed, "Fused delayed-scaling e4m3 quantization");
+  m.def("fp8_scale_update", &fp8_scale_update, "Fold accumulated amax into the fp8 scale");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

@@ -472,3 +472,70 @@ torch::Tensor add_rmsnorm_(torch::Tensor h, c10::optional<torch::Tensor> delta,
   HIP_CHECK_KERNEL();
   return y;
 }
+
+// ---------------------------------------------------------------------------
+// fp8 rollout: fused DELAYED-SCALING activation quantization.
+// Quantizes bf16 x with the scale computed from the PREVIOUS call's amax
+// (the standard fp8 delayed-scaling recipe) while accumulating this call's
+// amax — one pass, no fp32 materialization, no reduction dependency before
+// the GEMM. A 1-block epilogue kernel folds amax into the scale AFTER the
+// GEMM is enqueued (stream-ordered), ready for the next call.
+// ---------------------------------------------------------------------------
+
+#include <hip/hip_fp8.h>
+
+__global__ void fp8_quant_delayed_kernel(
+    const uint16_t* __restrict__ x,   // [n] bf16
+    uint8_t* __restrict__ y,          // [n] e4m3
+    const float* __restrict__ scale,  // [1] current scale (prev amax / 448)
+    float* __restrict__ amax_next,    // [1] running amax accumulator
+    int64_t n) {
+  const float inv_s = 1.f / scale[0];
+  float local_max = 0.f;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i0 < n;
+       i0 += (int64_t)gridDim.x * blockDim.x * 4) {
+    short4v xv = *reinterpret_cast<const short4v*>(x + i0);
+    uint8_t out[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = bf16_to_f32((uint16_t)xv[j]);
+      local_max = fmaxf(local_max, fabsf(v));
+      float q = fminf(fmaxf(v * inv_s, -448.f), 448.f);
+      out[j] = (uint8_t)__hip_cvt_float_to_fp8(q, __HIP_SATFINITE, __HIP_E4M3);
+    }
+    *reinterpret_cast<uint32_t*>(y + i0) = *reinterpret_cast<uint32_t*>(out);
+  }
+#pragma unroll
+  for (int off = 1; off < WAVE_SIZE; off <<= 1)
+    local_max = fmaxf(local_max, __shfl_xor(local_max, off, WAVE_SIZE));
+  if ((threadIdx.x & (WAVE_SIZE - 1)) == 0 && local_max > 0.f) {
+    atomicMax(reinterpret_cast<unsigned int*>(amax_next),
+              __float_as_uint(local_max));  // positive floats: uint order == float order
+  }
+}
+
+__global__ void fp8_scale_update_kernel(float* __restrict__ scale,
+                                        float* __restrict__ amax_next) {
+  const float a = amax_next[0];
+  scale[0] = fmaxf(a / 448.f, 1e-8f);
+  amax_next[0] = 0.f;
+}
+
+void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale,
+                       torch::Tensor amax_next) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(y.dtype() == torch::kFloat8_e4m3fn && y.numel() == x.numel());
+  TORCH_CHECK(scale.dtype() == torch::kFloat32 && amax_next.dtype() == torch::kFloat32);
+  const int64_t n = x.numel();
+  TORCH_CHECK(n % 4 == 0);
+  hipLaunchKernelGGL(fp8_quant_delayed_kernel, dim3(grid_for(n / 4, 256)), dim3(256), 0,
+                     cur_stream(), (const uint16_t*)x.data_ptr(), (uint8_t*)y.data_ptr(),
+                     scale.data_ptr<float>(), amax_next.data_ptr<float>(), n);
+  HIP_CHECK_KERNEL();
+}
+
+void fp8_scale_update(torch::Tensor scale, torch::Tensor amax_next) {
+  hipLaunchKernelGGL(fp8_scale_update_kernel, dim3(1), dim3(1), 0, cur_stream(),
+                     scale.data_ptr<float>(), amax_next.data_ptr<float>());
+  HIP_CHECK_KERNEL();
+}

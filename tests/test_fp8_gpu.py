@@ -79,3 +79,25 @@ def test_engine_fp8_decode_mode():
     b = feng.generate([prompts[0]], SamplingParams(temperature=0.0, max_tokens=8))[0]
     # greedy first token should agree on a sane quantization
     assert a.token_ids[0] == b.token_ids[0]
+
+
+@requires_gpu
+def test_fp8_linear_delayed_scaling_converges():
+    """Delayed scaling: call 1 uses the bootstrap scale; by call 2 the scale
+    reflects the true amax and outputs match the dynamic-quant path."""
+    M, N, K = 256, 2048, 1536
+    res = ops.pretune_fp8_decode_shapes([(M, N, K)], iters=5, verbose=False)
+    if not res:
+        pytest.skip("no valid fp8 algo")
+    w = torch.randn(N, K, device="cuda").to(torch.bfloat16)
+    w8, sw = ops.fp8_quant(w)
+    scale = torch.ones(1, device="cuda")
+    amax = torch.zeros(1, device="cuda")
+    x = (torch.randn(M, K, device="cuda") * 7).to(torch.bfloat16)
+    ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w)  # warms the scale
+    expect_scale = x.float().abs().amax() / 448.0
+    assert abs(scale.item() - expect_scale.item()) < 1e-3
+    y2 = ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w)
+    ref = ops.fp8_linear(x, w8, sw, w_bf16=w)
+    err = (y2.float() - ref.float()).abs().max().item()
+    assert err <= ref.float().abs().max().item() * 0.05 + 0.5, err
