@@ -282,25 +282,33 @@ class QwenModel(nn.Module):
         """(Re)quantize the four decode projections to OCP e4m3 with
         per-tensor scales. Call after every weight update (the engine's
         weight_version setter does) — the fp8 copies alias nothing."""
+        names = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj")
+        fresh = not hasattr(self, "_fp8_scales")
+        if fresh:
+            n_sites = len(self.layers) * len(names)
+            dev = next(self.parameters()).device
+            # every site's scale/amax is a VIEW into these, so ONE kernel per
+            # decode step folds all accumulated amaxes (fp8_scale_update_all)
+            self._fp8_scales = torch.ones(n_sites, device=dev, dtype=torch.float32)
+            self._fp8_amax = torch.zeros(n_sites, device=dev, dtype=torch.float32)
         with torch.no_grad():
+            site = 0
             for layer in self.layers:
                 old = getattr(layer, "_fp8", None)
                 q = {}
-                for name in ("qkv_proj", "o_proj", "gate_up_proj", "down_proj"):
+                for name in names:
                     w8, sw = ops.fp8_quant(getattr(layer, name).detach())
                     if old is not None and name in old:
-                        # keep the per-call-site activation-scale state (and
-                        # buffer identity — captured graphs hold its pointer)
-                        _, old_sw, scale_buf, amax_buf = old[name]
-                        old_w8 = old[name][0]
+                        # keep buffer identity — captured graphs hold pointers
+                        old_w8, old_sw, scale_v, amax_v = old[name]
                         old_w8.copy_(w8)
                         old_sw.copy_(sw)
-                        q[name] = (old_w8, old_sw, scale_buf, amax_buf)
+                        q[name] = (old_w8, old_sw, scale_v, amax_v)
                     else:
-                        dev = w8.device
                         q[name] = (w8, sw,
-                                   torch.ones(1, device=dev, dtype=torch.float32),
-                                   torch.zeros(1, device=dev, dtype=torch.float32))
+                                   self._fp8_scales[site : site + 1],
+                                   self._fp8_amax[site : site + 1])
+                    site += 1
                 layer._fp8 = q
         self.fp8_decode = True
 
@@ -403,6 +411,9 @@ class QwenModel(nn.Module):
     @torch.no_grad()
     def forward_decode(self, input_ids, positions, kv_cache, slot_mapping, block_tables, seq_lens):
         self._stamp_lora()
+        if getattr(self, "fp8_decode", False):
+            # fold last step's activation amaxes into this step's scales
+            ops.fp8_scale_update_all(self._fp8_scales, self._fp8_amax)
         h = self.embed_tokens[input_ids].contiguous()
         delta = None
         for i, layer in enumerate(self.layers):

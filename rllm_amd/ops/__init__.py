@@ -337,12 +337,14 @@ def fp8_linear(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
 
 def fp8_linear_delayed(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
                        scale_buf: torch.Tensor, amax_buf: torch.Tensor,
-                       w_bf16: torch.Tensor | None = None):
-    """Hot-path fp8 linear with DELAYED activation scaling: quantize with the
-    previous call's scale while accumulating this call's amax in the same
-    kernel; the scale folds forward after the GEMM (stream-ordered). Three
-    launches total (quant, GEMM, scale-update); graph-capture safe —
-    scale/amax are persistent per-call-site buffers."""
+                       w_bf16: torch.Tensor | None = None,
+                       update_scale: bool = False):
+    """Hot-path fp8 linear with DELAYED activation scaling: quantize with
+    this site's PREVIOUS scale while accumulating the new amax in the same
+    (native cvt_pk_fp8) kernel. Two launches (quant, GEMM); the scale fold
+    happens once per step via `fp8_scale_update_all` over every site's
+    view into a shared buffer (pass update_scale=True for standalone use).
+    Graph-capture safe — scale/amax are persistent per-site buffers."""
     C = require_ext()
     M, K = x.shape[0], x.shape[-1]
     N = w8.shape[0]
@@ -353,8 +355,15 @@ def fp8_linear_delayed(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
     x8 = torch.empty(x.shape, device=x.device, dtype=torch.float8_e4m3fn)
     C.fp8_quant_delayed(x.contiguous(), x8, scale_buf, amax_buf)
     y = C.hbl_fp8_mm(x8, w8, scale_buf, sw)
-    C.fp8_scale_update(scale_buf, amax_buf)
+    if update_scale:
+        C.fp8_scale_update(scale_buf, amax_buf)
     return y
+
+
+def fp8_scale_update_all(scales: torch.Tensor, amaxes: torch.Tensor):
+    """Fold every call site's accumulated amax into its scale (one launch;
+    called at the top of each decode step)."""
+    require_ext().fp8_scale_update(scales, amaxes)
 
 
 _FP8_TUNED: set = set()

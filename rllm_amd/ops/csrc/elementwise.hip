@@ -490,35 +490,50 @@ __global__ void fp8_quant_delayed_kernel(
     const float* __restrict__ scale,  // [1] current scale (prev amax / 448)
     float* __restrict__ amax_next,    // [1] running amax accumulator
     int64_t n) {
+  __shared__ float red[16];
   const float inv_s = 1.f / scale[0];
   float local_max = 0.f;
+  // 4 elements/thread/iter; native v_cvt_pk_fp8_f32 packs pairs
   for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i0 < n;
        i0 += (int64_t)gridDim.x * blockDim.x * 4) {
     short4v xv = *reinterpret_cast<const short4v*>(x + i0);
-    uint8_t out[4];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float v = bf16_to_f32((uint16_t)xv[j]);
-      local_max = fmaxf(local_max, fabsf(v));
-      float q = fminf(fmaxf(v * inv_s, -448.f), 448.f);
-      out[j] = (uint8_t)__hip_cvt_float_to_fp8(q, __HIP_SATFINITE, __HIP_E4M3);
-    }
-    *reinterpret_cast<uint32_t*>(y + i0) = *reinterpret_cast<uint32_t*>(out);
+    float v0 = bf16_to_f32((uint16_t)xv[0]);
+    float v1 = bf16_to_f32((uint16_t)xv[1]);
+    float v2 = bf16_to_f32((uint16_t)xv[2]);
+    float v3 = bf16_to_f32((uint16_t)xv[3]);
+    local_max = fmaxf(local_max, fmaxf(fmaxf(fabsf(v0), fabsf(v1)), fmaxf(fabsf(v2), fabsf(v3))));
+    // clamp to e4m3 max-normal (448): OCP e4m3 has no inf to saturate to
+    auto cl = [](float v) { return fminf(fmaxf(v, -448.f), 448.f); };
+    int packed = 0;
+    packed = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v0 * inv_s), cl(v1 * inv_s), packed, false);
+    packed = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v2 * inv_s), cl(v3 * inv_s), packed, true);
+    *reinterpret_cast<uint32_t*>(y + i0) = (uint32_t)packed;
   }
+  // wave max -> LDS -> block max -> ONE atomic per block (same-address
+  // atomics from every wave serialize badly)
 #pragma unroll
   for (int off = 1; off < WAVE_SIZE; off <<= 1)
     local_max = fmaxf(local_max, __shfl_xor(local_max, off, WAVE_SIZE));
-  if ((threadIdx.x & (WAVE_SIZE - 1)) == 0 && local_max > 0.f) {
-    atomicMax(reinterpret_cast<unsigned int*>(amax_next),
-              __float_as_uint(local_max));  // positive floats: uint order == float order
+  const int wid = threadIdx.x / WAVE_SIZE;
+  if ((threadIdx.x & (WAVE_SIZE - 1)) == 0) red[wid] = local_max;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = red[0];
+    for (int w = 1; w < (int)(blockDim.x / WAVE_SIZE); ++w) m = fmaxf(m, red[w]);
+    if (m > 0.f)
+      atomicMax(reinterpret_cast<unsigned int*>(amax_next), __float_as_uint(m));
   }
 }
 
+// fold accumulated amax into scales for n call sites (one launch per
+// decode step, not per GEMM)
 __global__ void fp8_scale_update_kernel(float* __restrict__ scale,
-                                        float* __restrict__ amax_next) {
-  const float a = amax_next[0];
-  scale[0] = fmaxf(a / 448.f, 1e-8f);
-  amax_next[0] = 0.f;
+                                        float* __restrict__ amax_next, int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const float a = amax_next[i];
+  if (a > 0.f) scale[i] = fmaxf(a / 448.f, 1e-8f);
+  amax_next[i] = 0.f;
 }
 
 void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale,
@@ -535,7 +550,9 @@ void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale,
 }
 
 void fp8_scale_update(torch::Tensor scale, torch::Tensor amax_next) {
-  hipLaunchKernelGGL(fp8_scale_update_kernel, dim3(1), dim3(1), 0, cur_stream(),
-                     scale.data_ptr<float>(), amax_next.data_ptr<float>());
+  const int n = (int)scale.numel();
+  TORCH_CHECK(amax_next.numel() == n && scale.is_contiguous() && amax_next.is_contiguous());
+  hipLaunchKernelGGL(fp8_scale_update_kernel, dim3((n + 255) / 256), dim3(256), 0, cur_stream(),
+                     scale.data_ptr<float>(), amax_next.data_ptr<float>(), n);
   HIP_CHECK_KERNEL();
 }

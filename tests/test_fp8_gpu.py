@@ -86,18 +86,19 @@ def test_fp8_linear_delayed_scaling_converges():
     """Delayed scaling: call 1 uses the bootstrap scale; by call 2 the scale
     reflects the true amax and outputs match the dynamic-quant path."""
     M, N, K = 256, 2048, 1536
-    res = ops.pretune_fp8_decode_shapes([(M, N, K)], iters=5, verbose=False)
-    if not res:
+    ops.pretune_fp8_decode_shapes([(M, N, K)], iters=5, verbose=False)
+    if not ops.require_ext().hbl_fp8_has(M, N, K):
         pytest.skip("no valid fp8 algo")
     w = torch.randn(N, K, device="cuda").to(torch.bfloat16)
     w8, sw = ops.fp8_quant(w)
     scale = torch.ones(1, device="cuda")
     amax = torch.zeros(1, device="cuda")
     x = (torch.randn(M, K, device="cuda") * 7).to(torch.bfloat16)
-    ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w)  # warms the scale
+    # warms the scale (explicit per-call update in standalone use)
+    ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w, update_scale=True)
     expect_scale = x.float().abs().amax() / 448.0
-    assert abs(scale.item() - expect_scale.item()) < 1e-3
-    y2 = ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w)
+    assert abs(scale.item() - expect_scale.item()) < 2e-3 * max(1.0, expect_scale.item())
+    y2 = ops.fp8_linear_delayed(x, w8, sw, scale, amax, w_bf16=w, update_scale=True)
     ref = ops.fp8_linear(x, w8, sw, w_bf16=w)
     err = (y2.float() - ref.float()).abs().max().item()
     assert err <= ref.float().abs().max().item() * 0.05 + 0.5, err
