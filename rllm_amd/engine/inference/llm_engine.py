@@ -591,10 +591,12 @@ class LLMEngine:
         seq.state = SeqState.FINISHED
         seq.finish_reason = reason
         if seq.pages:
-            if self.prefix_cache is not None and reason in ("stop", "length"):
+            if (self.prefix_cache is not None and reason in ("stop", "length")
+                    and seq.weight_version == self._weight_version):
                 # make this sequence's full pages (prompt + generated)
                 # available to future requests — multi-turn rollouts extend
-                # exactly this token prefix
+                # exactly this token prefix. Skipped when the weights moved
+                # mid-generation (async training): that KV is stale.
                 self.prefix_cache.publish(seq.prompt_ids + seq.output_ids,
                                           seq.pages, seq.n_cached_pages)
             self._release_pages(seq.pages)

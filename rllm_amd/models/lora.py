@@ -106,6 +106,8 @@ def merge_lora_(model) -> None:
             delta = (lora[f"{name}_B"].float() @ lora[f"{name}_A"].float()) * layer.lora_scale
             w.data.copy_((w.data.float() + delta).to(w.dtype))
     model.lora_merged = True
+    if getattr(model, "fp8_decode", False):
+        model.enable_fp8_decode()  # re-quantize: merge mutated the base weights
 
 
 @torch.no_grad()
@@ -121,6 +123,8 @@ def unmerge_lora_(model) -> None:
             delta = (lora[f"{name}_B"].float() @ lora[f"{name}_A"].float()) * layer.lora_scale
             w.data.copy_((w.data.float() - delta).to(w.dtype))
     model.lora_merged = False
+    if getattr(model, "fp8_decode", False):
+        model.enable_fp8_decode()
 
 
 @contextmanager
