@@ -20,6 +20,7 @@ class LazyGroup(click.Group):
         "eval": "rllm_amd.cli.eval_cmd:eval_cmd",
         "dataset": "rllm_amd.cli.dataset_cmd:dataset_cmd",
         "view": "rllm_amd.cli.view:view",
+        "serve": "rllm_amd.cli.serve:serve",
     }
 
     def list_commands(self, ctx):

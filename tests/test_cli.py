@@ -59,3 +59,17 @@ def test_train_cpu_smoke(tmp_path):
         "--max-steps", "1", "--registry-dir", reg, "--logger", "console",
     ])
     assert r.exit_code == 0, r.output
+
+
+def test_serve_help_and_gpu_guard():
+    from click.testing import CliRunner
+
+    from rllm_amd.cli.main import cli
+
+    r = CliRunner().invoke(cli, ["serve", "--help"])
+    assert r.exit_code == 0 and "OpenAI-compatible" in r.output
+    import torch
+
+    if not torch.cuda.is_available():
+        r2 = CliRunner().invoke(cli, ["serve"])
+        assert r2.exit_code != 0  # clean error, not a crash
