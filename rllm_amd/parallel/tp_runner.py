@@ -98,9 +98,7 @@ class TPLockstepEngine:
         """Blocking batch generation through the lockstep loop (bench/tests);
         mirrors LLMEngine.generate."""
         assert self.is_driver
-        from rllm_amd.engine.inference.llm_engine import SamplingParams
-
-        if isinstance(params, SamplingParams):
+        if not isinstance(params, (list, tuple)):
             params = [params] * len(prompts)
         for i, (p, sp) in enumerate(zip(prompts, params)):
             self.add_request(f"gen-{i}", p, sp)

@@ -307,3 +307,48 @@ def test_tp_lockstep_world2():
             p.terminate()
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def _worker_tp_generate(rank, world, q):
+    try:
+        pdist = _setup_dist(rank, world)
+        import torch.distributed as dist
+
+        from rllm_amd.parallel.tp_runner import TPLockstepEngine
+
+        eng = _FakeEngine()
+        wrap = TPLockstepEngine(eng, tp_group=dist.group.WORLD)
+        if wrap.is_driver:
+            outs = wrap.generate([[1, 2], [9]], params=2)
+            assert [o.token_ids for o in outs] == [[3, 4], [9, 10]], outs
+        else:
+            wrap.serve()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, eng.log)
+        assert gathered[0] == gathered[1]
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, f"FAIL: {traceback.format_exc()[-400:]}"))
+
+
+def test_tp_generate_world2():
+    import os
+
+    import torch.multiprocessing as mp
+
+    os.environ["TEST_DIST_PORT"] = "29629"
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_tp_generate, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(), q.get()]
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
