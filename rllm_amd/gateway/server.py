@@ -174,10 +174,13 @@ def create_app(config: GatewayConfig | None = None,
                                 add_return_token_ids=st.config.add_return_token_ids)
 
         # cumulative token mode: rewrite turn>=2 chat requests to a
-        # pre-tokenized completions call (prefix-extension invariant)
+        # pre-tokenized completions call (prefix-extension invariant).
+        # Streaming requests are supported on the colocated local_handler
+        # path (the SSE events are synthesized from the finished response).
         cumulative = False
         if (st.accumulator is not None and sid is not None
-                and path.endswith("chat/completions") and not mutated.get("stream")):
+                and path.endswith("chat/completions")
+                and (not mutated.get("stream") or st.proxy.local_handler is not None)):
             prompt_ids = st.accumulator.build_prompt_ids(sid, body.get("messages") or [])
             if prompt_ids is not None:
                 mutated["prompt_token_ids"] = prompt_ids
@@ -210,6 +213,12 @@ def create_app(config: GatewayConfig | None = None,
                     if trace.weight_version is None:
                         trace.weight_version = st.weight_version
                     await st.store.add_trace(trace)
+                    if (st.accumulator is not None and sid is not None
+                            and (cumulative or body.get("messages"))
+                            and trace.completion_token_ids):
+                        st.accumulator.record_turn(sid, body.get("messages") or [],
+                                                   trace.prompt_token_ids,
+                                                   trace.completion_token_ids)
 
             return StreamingResponse(gen(), media_type="text/event-stream")
 

@@ -236,3 +236,45 @@ def test_streaming_with_local_handler():
         client.delete_session("st:0")
     finally:
         gw.stop()
+
+
+def test_cumulative_streaming_local_handler():
+    """Cumulative token mode over SSE (colocated path): turn 2 streamed chat
+    still becomes a pre-tokenized prefix-extension completions call."""
+    from rllm_amd.gateway.native_adapter import make_torch_lm_local_handler
+    from rllm_amd.models.torch_lm import TinyTorchLM
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    handler = make_torch_lm_local_handler(TinyTorchLM(seed=0), parser)
+    gw = GatewayManager(GatewayConfig(cumulative_token_mode=True),
+                        local_handler=handler, parser=parser)
+    gw.start()
+    try:
+        client = gw.client()
+        client.create_session("cs:0", sampling_params={"max_tokens": 4})
+        url = gw.session_url("cs:0") + "/chat/completions"
+        msgs = [{"role": "user", "content": "hi"}]
+
+        def stream_turn(m):
+            with httpx.stream("POST", url, json={"model": "m", "stream": True,
+                                                 "messages": m}, timeout=60.0) as r:
+                lines = [ln for ln in r.iter_lines() if ln.startswith("data:")]
+            assert lines[-1].strip() == "data: [DONE]"
+
+        stream_turn(msgs)
+        t1 = client.get_traces("cs:0")[0]
+        reply1 = t1.response_message["content"]
+        msgs2 = msgs + [{"role": "assistant", "content": reply1},
+                        {"role": "user", "content": "again"}]
+        stream_turn(msgs2)
+        traces = client.get_traces("cs:0")
+        assert len(traces) == 2
+        t1, t2 = traces
+        expected_prefix = t1.prompt_token_ids + t1.completion_token_ids
+        assert t2.prompt_token_ids[: len(expected_prefix)] == expected_prefix
+        assert len(t2.prompt_token_ids) > len(expected_prefix)
+        client.delete_session("cs:0")
+    finally:
+        gw.stop()
