@@ -42,28 +42,40 @@ def is_initialized() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
+# Default process group for ALL helpers below. None = WORLD (the usual DP
+# case). Separated placement sets this to the trainer subgroup so
+# PolicyTrainer's collectives never involve rollout ranks (which sit in
+# their own command loop).
+_DEFAULT_GROUP = None
+
+
+def set_default_group(group) -> None:
+    global _DEFAULT_GROUP
+    _DEFAULT_GROUP = group
+
+
 def get_rank() -> int:
-    return dist.get_rank() if is_initialized() else 0
+    return dist.get_rank(group=_DEFAULT_GROUP) if is_initialized() else 0
 
 
 def get_world_size() -> int:
-    return dist.get_world_size() if is_initialized() else 1
+    return dist.get_world_size(group=_DEFAULT_GROUP) if is_initialized() else 1
 
 
 def barrier():
     if is_initialized():
-        dist.barrier()
+        dist.barrier(group=_DEFAULT_GROUP)
 
 
 def all_reduce_sum_(tensor: torch.Tensor) -> torch.Tensor:
     if is_initialized():
-        dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+        dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=_DEFAULT_GROUP)
     return tensor
 
 
 def all_reduce_max_(tensor: torch.Tensor) -> torch.Tensor:
     if is_initialized():
-        dist.all_reduce(tensor, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=_DEFAULT_GROUP)
     return tensor
 
 
@@ -73,13 +85,14 @@ def all_reduce_scalar(value: float, op: str = "sum") -> float:
         return value
     device = "cuda" if dist.get_backend() == "nccl" else "cpu"
     t = torch.tensor([value], dtype=torch.float64, device=device)
-    dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM,
+                    group=_DEFAULT_GROUP)
     return float(t.item())
 
 
 def broadcast_(tensor: torch.Tensor, src: int = 0) -> torch.Tensor:
     if is_initialized():
-        dist.broadcast(tensor, src=src)
+        dist.broadcast(tensor, src=src, group=_DEFAULT_GROUP)
     return tensor
 
 
@@ -87,8 +100,8 @@ def all_gather_object_list(obj) -> list:
     """C4: gather python objects (episodes/trajectories) from all ranks."""
     if not is_initialized():
         return [obj]
-    out = [None] * dist.get_world_size()
-    dist.all_gather_object(out, obj)
+    out = [None] * dist.get_world_size(group=_DEFAULT_GROUP)
+    dist.all_gather_object(out, obj, group=_DEFAULT_GROUP)
     return out
 
 

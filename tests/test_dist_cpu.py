@@ -202,3 +202,33 @@ def _worker_separated_topology(rank, world, q):
 @pytest.mark.timeout(180)
 def test_separated_topology_world2():
     _run_spawn(_worker_separated_topology, "29617")
+
+
+def _worker_default_group(rank, world, q):
+    """Group-scoped helpers: with the default group set to a singleton
+    subgroup, collectives stay inside it (separated-placement guarantee)."""
+    try:
+        pdist = _setup(rank, world)
+        import torch.distributed as dist
+
+        g0 = dist.new_group([0])
+        g1 = dist.new_group([1])
+        pdist.set_default_group(g0 if rank == 0 else g1)
+        assert pdist.get_world_size() == 1
+        assert pdist.get_rank() == 0
+        # sums see only this rank's contribution
+        assert pdist.all_reduce_scalar(float(rank + 5)) == rank + 5
+        objs = pdist.all_gather_object_list({"r": rank})
+        assert objs == [{"r": rank}]
+        pdist.set_default_group(None)
+        assert pdist.get_world_size() == 2
+        pdist.barrier()
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+@pytest.mark.timeout(180)
+def test_default_group_scoping_world2():
+    _run_spawn(_worker_default_group, "29619")
