@@ -186,3 +186,25 @@ def test_tis_weights_token_and_sequence():
         tis_weights(old, roll, mode="sequence", cap=1.0)
     with pytest.raises(ValueError):
         tis_weights(old, roll, mode="banana", cap=1.0)
+
+
+def test_advantage_properties_random():
+    """Estimator invariants over random groups: GRPO advantages are
+    mean-zero per group; RLOO baselines are leave-one-out means;
+    uniform-reward groups give (near-)zero advantage."""
+    import random
+
+    rng = random.Random(0)
+    for trial in range(50):
+        n = rng.randint(2, 9)
+        r = np.array([rng.choice([0.0, 0.5, 1.0]) for _ in range(n)])
+        grpo, _ = calculate_grpo_advantages_per_group(r)
+        assert abs(float(np.sum(grpo))) < 1e-4 * n
+        rloo, _ = calculate_rloo_advantages_per_group(r)
+        for i in range(n):
+            others = np.delete(r, i)
+            assert abs(float(rloo[i]) - (r[i] - float(others.mean()))) < 1e-6
+        uniform = np.full(n, 0.7)
+        g, _ = calculate_grpo_advantages_per_group(uniform)
+        l, _ = calculate_rloo_advantages_per_group(uniform)
+        assert np.all(np.abs(g) < 1e-4) and np.all(np.abs(l) < 1e-6)
