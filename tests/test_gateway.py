@@ -278,3 +278,43 @@ def test_cumulative_streaming_local_handler():
         client.delete_session("cs:0")
     finally:
         gw.stop()
+
+
+def test_token_accumulator_invariant_random_turns():
+    """Prefix-extension invariant over randomized multi-turn sessions:
+    every continuation's prompt ids start with the previous turn's
+    prompt+completion ids, for arbitrary message contents/counts."""
+    import random
+
+    from rllm_amd.gateway.token_accumulator import TokenAccumulator
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    rng = random.Random(3)
+    for trial in range(20):
+        acc = TokenAccumulator(parser)
+        sid = f"s{trial}"
+        msgs = [{"role": "user", "content": f"q{trial}-" + "x" * rng.randint(1, 20)}]
+        assert acc.build_prompt_ids(sid, msgs) is None  # first turn: plain path
+        prompt_ids = parser.tokenizer.encode(
+            "".join(parser.format_message(m) for m in msgs) + parser.assistant_prefix)
+        prev = None
+        for turn in range(rng.randint(2, 5)):
+            completion_ids = [rng.randrange(40, 120) for _ in range(rng.randint(1, 8))]
+            acc.record_turn(sid, msgs, prompt_ids, completion_ids)
+            expected_prefix = list(prompt_ids) + completion_ids
+            reply = "".join(chr(c) for c in completion_ids)
+            n_new = rng.randint(1, 2)
+            msgs = msgs + [{"role": "assistant", "content": reply}] + [
+                {"role": "user", "content": f"t{turn}n{k}-" + "y" * rng.randint(0, 9)}
+                for k in range(n_new)]
+            nxt = acc.build_prompt_ids(sid, msgs)
+            assert nxt is not None
+            assert nxt[: len(expected_prefix)] == expected_prefix
+            assert len(nxt) > len(expected_prefix)  # the rendered delta
+            prev, prompt_ids = expected_prefix, nxt
+        # shrunken history is NOT a continuation
+        assert acc.build_prompt_ids(sid, msgs[:1]) is None
+        acc.reset(sid)
+        assert acc.build_prompt_ids(sid, msgs) is None
