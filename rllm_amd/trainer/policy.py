@@ -148,6 +148,10 @@ class PolicyTrainer:
         n_local_tokens = sum(sum(r.response_mask) for r in rows)
         n_global_tokens = pdist.all_reduce_scalar(float(n_local_tokens), op="sum")
         world = pdist.get_world_size()
+        # seq-mean denominator: ONE collective, BEFORE the micro loop — ranks
+        # have different micro counts, so any collective inside it deadlocks
+        n_global_rows = (pdist.all_reduce_scalar(float(len(rows)), op="sum")
+                         if cfg.loss_agg_mode == "seq-mean-token-mean" else 0.0)
         if n_global_tokens <= 0:
             logger.warning("update_policy called with zero response tokens; skipping step")
             return {"actor/skipped": 1.0}
@@ -213,8 +217,8 @@ class PolicyTrainer:
                 # global-token-mean: SUM-all-reduce of grads then NO rescale
                 loss = loss_tok.sum() / n_global_tokens
             elif cfg.loss_agg_mode == "seq-mean-token-mean":
-                # per-sequence token mean, then mean over the global row count
-                n_global_rows = pdist.all_reduce_scalar(float(batch.n_rows), op="sum") if world > 1 else batch.n_rows
+                # per-sequence token mean, then mean over the GLOBAL sequence
+                # count (computed once above)
                 loss = 0.0
                 for s in seq_ids.unique():
                     m = seq_ids == s
