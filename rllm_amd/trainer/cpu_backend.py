@@ -30,7 +30,8 @@ class CPUBackend(BackendProtocol):
     def __init__(self, agent_flow, evaluator=None, model: TinyTorchLM | None = None,
                  lr: float = 1e-4, eps_clip: float = 0.2, kl_beta: float = 0.0,
                  n_parallel_tasks: int = 16, rollout_max_tokens: int = 16,
-                 temperature: float = 1.0, seed: int = 0):
+                 temperature: float = 1.0, seed: int = 0,
+                 gateway_config: GatewayConfig | None = None):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
         self.model = model or TinyTorchLM(seed=seed)
@@ -52,6 +53,7 @@ class CPUBackend(BackendProtocol):
         self.gateway: GatewayManager | None = None
         self.flow_engine: AgentFlowEngine | None = None
         self.weight_version_ref = {"v": 0}
+        self.gateway_config = gateway_config
 
     # ------------------------------------------------------------------
     def init_rollout_engine(self):
@@ -60,7 +62,10 @@ class CPUBackend(BackendProtocol):
         handler = make_torch_lm_local_handler(
             self.model, self.parser, eos_token_id=None, seed=self.seed,
             weight_version_ref=self.weight_version_ref)
-        self.gateway = GatewayManager(GatewayConfig(), local_handler=handler)
+        gw_cfg = self.gateway_config or GatewayConfig()
+        self.gateway = GatewayManager(
+            gw_cfg, local_handler=handler,
+            parser=self.parser if gw_cfg.cumulative_token_mode else None)
         self.gateway.start()
         self.flow_engine = AgentFlowEngine(
             self.agent_flow, self.gateway, evaluator=self.evaluator,

@@ -101,3 +101,43 @@ def test_trainer_validation(tmp_path):
     trainer.fit()
     # validation ran without blowing up; metrics tracked on state
     assert trainer.state.global_step == 1
+
+
+def test_cpu_backend_cumulative_multiturn(tmp_path):
+    """Config-1 shape WITH cumulative token mode: whole pipeline on CPU,
+    two-turn flow, merged packed rows (prefix-extension end-to-end with no
+    GPU in the loop)."""
+    import httpx
+
+    import rllm_amd
+    from rllm_amd.data.dataset import Dataset
+    from rllm_amd.gateway.models import GatewayConfig
+    from rllm_amd.trainer.cpu_backend import CPUBackend
+    from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+
+    @rllm_amd.rollout
+    def two_turn(task, config):
+        msgs = [{"role": "user", "content": str(task.instruction)}]
+        r1 = httpx.post(config.base_url + "/chat/completions",
+                        json={"model": config.model, "messages": msgs, "max_tokens": 4},
+                        timeout=60.0)
+        r1.raise_for_status()
+        reply = r1.json()["choices"][0]["message"]["content"]
+        msgs += [{"role": "assistant", "content": reply},
+                 {"role": "user", "content": "and?"}]
+        httpx.post(config.base_url + "/chat/completions",
+                   json={"model": config.model, "messages": msgs, "max_tokens": 4},
+                   timeout=60.0).raise_for_status()
+
+    @rllm_amd.evaluator
+    def ev(task, episode):
+        return 1.0 if episode.trajectories and episode.trajectories[0].steps else 0.0
+
+    backend = CPUBackend(two_turn, ev, n_parallel_tasks=4, rollout_max_tokens=4,
+                         gateway_config=GatewayConfig(cumulative_token_mode=True))
+    tasks = Dataset([{"question": f"q{i}", "id": str(i)} for i in range(2)]).as_tasks(id_key="id")
+    tcfg = TrainerConfig(total_epochs=1, train_batch_size=2, rollout_n=2, max_steps=1,
+                         logger_backends=[])
+    trainer = UnifiedTrainer(backend, tasks, config=tcfg)
+    trainer.fit()
+    assert trainer.state.global_step == 1
