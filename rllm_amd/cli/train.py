@@ -38,9 +38,11 @@ def resolve_ref(ref: str):
 @click.option("--save-freq", default=0)
 @click.option("--registry-dir", default=None)
 @click.option("--logger", "loggers", multiple=True, default=["console"])
+@click.option("--lora-rank", default=0, type=int, help="train LoRA adapters of this rank (0 = full finetune)")
+@click.option("--cumulative", is_flag=True, help="gateway cumulative token mode (multi-turn agents)")
 def train(dataset, agent_ref, evaluator_ref, backend, split, epochs, batch_size, rollout_n,
           max_steps, lr, kl_beta, estimator, model_config, checkpoint_dir, save_freq,
-          registry_dir, loggers):
+          registry_dir, loggers, lora_rank, cumulative):
     """Train an agent with GRPO/RLOO on a registered dataset."""
     from rllm_amd.data.dataset import DatasetRegistry
     from rllm_amd.trainer.agent_trainer import AgentTrainer
@@ -55,6 +57,14 @@ def train(dataset, agent_ref, evaluator_ref, backend, split, epochs, batch_size,
     backend_kwargs = {}
     if backend == "native":
         backend_kwargs["model_config"] = model_config
+        if lora_rank > 0:
+            from rllm_amd.models.lora import LoRAConfig
+
+            backend_kwargs["lora"] = LoRAConfig(r=lora_rank, alpha=2 * lora_rank)
+    if cumulative:
+        from rllm_amd.gateway.models import GatewayConfig
+
+        backend_kwargs["gateway_config"] = GatewayConfig(cumulative_token_mode=True)
 
     cfg = TrainerConfig(
         total_epochs=epochs, train_batch_size=batch_size, rollout_n=rollout_n,
