@@ -61,6 +61,7 @@ class NativeBackend(BackendProtocol):
         seed: int = 0,
         lora=None,  # models.lora.LoRAConfig | True for defaults | None = full finetune
         gateway_config=None,  # GatewayConfig; e.g. cumulative_token_mode=True for multi-turn
+        fp8_rollout: bool = False,  # e4m3 decode GEMMs (wins at >=14B; profiles/)
     ):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
@@ -80,6 +81,7 @@ class NativeBackend(BackendProtocol):
         self.checkpoint_path = checkpoint_path
         self.lora_config = lora
         self.gateway_config = gateway_config
+        self.fp8_rollout = fp8_rollout
 
         self.model: QwenModel | None = None
         self.ref_model: QwenModel | None = None
@@ -131,7 +133,8 @@ class NativeBackend(BackendProtocol):
         self.engine = LLMEngine(self.model, kv_budget_bytes=self.kv_budget_bytes,
                                 max_num_seqs=self.max_num_seqs,
                                 max_num_batched_tokens=self.max_num_batched_tokens,
-                                eos_token_id=eos, seed=self.seed)
+                                eos_token_id=eos, seed=self.seed,
+                                fp8_decode=self.fp8_rollout or None)
         self.driver = AsyncEngineDriver(self.engine)
         self.rollout_engine = NativeEngine(self.driver, self.parser,
                                            max_prompt_length=self.max_prompt_length,
