@@ -59,6 +59,9 @@ class TrainerConfig:
     experiment_name: str = "run"
     episode_log_dir: str | None = None
     seed: int = 0
+    # rocprofv3-visible step gating (reference trainer.profile_steps):
+    # wrap the listed global steps in hipProfilerStart/Stop + roctx ranges
+    profile_steps: list | None = None
 
 
 @dataclass
@@ -137,7 +140,10 @@ class UnifiedTrainer:
                         return
                     t0 = time.monotonic()
                     self.backend.on_batch_start(self.state.global_step)
-                    metrics = await self._train_batch_async(task_batch)
+                    from rllm_amd.utils.profiling import step_profile_region
+
+                    with step_profile_region(self.state.global_step, cfg.profile_steps):
+                        metrics = await self._train_batch_async(task_batch)
                     metrics["time/step_s"] = time.monotonic() - t0
                     metrics["epoch"] = epoch
                     self.tracking.log(metrics, step=self.state.global_step)

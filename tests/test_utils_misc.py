@@ -121,3 +121,14 @@ def test_split_thought():
     assert th == "hmm 2+2" and ans == "4"
     th2, ans2 = split_thought("just 4")
     assert th2 == "" and ans2 == "just 4"
+
+
+def test_step_profile_region_cpu_noop():
+    from rllm_amd.utils.profiling import roctx_range, step_profile_region
+
+    with step_profile_region(3, [3]) as active:
+        pass  # no GPU here: must be a no-op, not an error
+    with step_profile_region(2, [3]) as active2:
+        assert active2 is False
+    with roctx_range("x"):
+        pass
