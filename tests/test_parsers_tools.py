@@ -108,3 +108,44 @@ def test_code_reward_assert_style_and_extraction():
 def test_code_reward_timeout():
     out = run_tests("while True: pass", [{"assert": "assert True"}], timeout=1.5)
     assert not out.is_correct and out.error == "timeout"
+
+
+def test_qwen_parser_matches_hf_jinja_template():
+    """Render-equivalence vs the real Qwen2 Jinja chat template executed by
+    transformers' apply_chat_template (offline: a minimal fast tokenizer
+    object carries the template; only rendered STRINGS are compared)."""
+    try:
+        from tokenizers import Tokenizer, models
+        from transformers import PreTrainedTokenizerFast
+    except ImportError:
+        import pytest
+
+        pytest.skip("transformers/tokenizers unavailable")
+
+    QWEN2_TEMPLATE = (
+        "{% for message in messages %}"
+        "{{'<|im_start|>' + message['role'] + '\n' + message['content'] + '<|im_end|>' + '\n'}}"
+        "{% endfor %}"
+        "{% if add_generation_prompt %}{{ '<|im_start|>assistant\n' }}{% endif %}"
+    )
+    hf_tok = PreTrainedTokenizerFast(
+        tokenizer_object=Tokenizer(models.WordLevel({"[UNK]": 0}, unk_token="[UNK]")),
+        chat_template=QWEN2_TEMPLATE)
+
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    for msgs in (
+        [{"role": "system", "content": "be brief"},
+         {"role": "user", "content": "hi there"}],
+        [{"role": "user", "content": "q1"},
+         {"role": "assistant", "content": "a1"},
+         {"role": "user", "content": "q2"}],
+    ):
+        ours = parser.format(msgs, add_generation_prompt=True)
+        theirs = hf_tok.apply_chat_template(msgs, tokenize=False, add_generation_prompt=True)
+        assert ours == theirs, (ours, theirs)
+        ours_nogen = parser.format(msgs, add_generation_prompt=False)
+        theirs_nogen = hf_tok.apply_chat_template(msgs, tokenize=False, add_generation_prompt=False)
+        assert ours_nogen == theirs_nogen
