@@ -77,6 +77,10 @@ class NativeBackend(BackendProtocol):
         self.max_num_batched_tokens = max_num_batched_tokens
         self.max_prompt_length = max_prompt_length
         self.rollout_sampling_params = rollout_sampling_params or {"temperature": 1.0, "top_p": 1.0}
+        # rollout temperature must flow into every training-side logprob
+        # (new/old/ref) or the IS ratios compare mismatched distributions
+        # (reference verl_backend.py:612 meta_info['temperature'])
+        self.policy_config.temperature = float(self.rollout_sampling_params.get("temperature", 1.0))
         self.n_parallel_tasks = n_parallel_tasks
         self.checkpoint_path = checkpoint_path
         self.lora_config = lora
@@ -166,11 +170,12 @@ class NativeBackend(BackendProtocol):
             with torch.no_grad():
                 hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
                 lp, _ = ops.chunked_logprob(hidden[rows_idx], self.model.lm_weight,
-                                            batch.targets[rows_idx], want_entropy=False)
+                                            batch.targets[rows_idx],
+                                            temperature=self.policy_config.temperature,
+                                            want_entropy=False)
                 return lp
 
-        return self.policy.update_policy(
-            rows, old_logprob_fn=None if self.policy_config.bypass_mode else old_lp_fn)
+        return self.policy.update_policy(rows, old_logprob_fn=old_lp_fn)
 
     def on_policy_updated(self, weight_version: int) -> None:
         # colocated: engine shares the actor's tensors — only the version
@@ -192,7 +197,7 @@ class NativeBackend(BackendProtocol):
     def load_checkpoint(self, path: str) -> int:
         sd = torch.load(f"{path}/actor.pt", weights_only=True, map_location=self.device)
         self.policy.load_state_dict(sd)
-        return 0
+        return int(sd.get("weight_version", 0))
 
     def shutdown(self) -> None:
         if self.gateway is not None:

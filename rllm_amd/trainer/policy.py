@@ -50,6 +50,12 @@ class PolicyTrainerConfig:
     # (deterministic kernels, same weights) but saves a full forward pass;
     # "rollout" = behaves like bypass_mode.
     old_logprob_mode: str = "recompute"
+    # rollout sampling temperature: the sampler records logprobs of
+    # logits/T (sampling.hip), so every training-side logprob (new, old,
+    # ref) must be computed at the SAME temperature or the IS ratios and
+    # KL compare mismatched distributions (reference propagates it via
+    # meta_info['temperature'], verl_backend.py:612).
+    temperature: float = 1.0
     entropy_chunk: int = 16384
     use_ref: bool = True
     # LoRA training: the KL reference is the actor's own base weights with
@@ -123,7 +129,9 @@ class PolicyTrainer:
         h = hidden[rows]
         tgt = batch.targets[rows]
         lp, ent = ops.chunked_logprob(h, model.lm_weight, tgt,
-                                      chunk=self.cfg.entropy_chunk, want_entropy=want_entropy)
+                                      chunk=self.cfg.entropy_chunk,
+                                      temperature=self.cfg.temperature,
+                                      want_entropy=want_entropy)
         return lp, (ent.mean().item() if ent is not None else None)
 
     # ------------------------------------------------------------------
@@ -174,16 +182,24 @@ class PolicyTrainer:
             rollout_lp = batch.rollout_logprobs[rows_idx]
 
             lp, ent = ops.chunked_logprob(h, self.model.lm_weight, tgt,
-                                          chunk=cfg.entropy_chunk, want_entropy=True)
+                                          chunk=cfg.entropy_chunk,
+                                          temperature=cfg.temperature, want_entropy=True)
             if ent is not None:
                 ent_sum += float(ent.sum())
                 ent_n += ent.numel()
 
             with torch.no_grad():
-                if cfg.bypass_mode or cfg.old_logprob_mode == "rollout" or old_logprob_fn is None:
+                if cfg.bypass_mode or cfg.old_logprob_mode == "rollout":
                     old_lp = rollout_lp
                 elif cfg.old_logprob_mode == "alias":
                     old_lp = lp.detach()
+                elif old_logprob_fn is None:
+                    # recompute mode REQUIRES the callback — silently falling
+                    # back to rollout logprobs would give bypass semantics
+                    # without the caller asking for them
+                    raise ValueError(
+                        "old_logprob_mode='recompute' needs old_logprob_fn; pass one "
+                        "or set old_logprob_mode='rollout'/bypass_mode=True explicitly")
                 else:
                     old_lp = old_logprob_fn(batch, rows_idx)
                 ref_lp = None
@@ -193,15 +209,21 @@ class PolicyTrainer:
                     with _lora.disabled(self.model):
                         ref_hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
                         ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.model.lm_weight,
-                                                        tgt, chunk=cfg.entropy_chunk, want_entropy=False)
+                                                        tgt, chunk=cfg.entropy_chunk,
+                                                        temperature=cfg.temperature, want_entropy=False)
                 elif cfg.use_ref and self.ref_model is not None:
                     ref_hidden = self.ref_model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
                     ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.ref_model.lm_weight,
-                                                    tgt, chunk=cfg.entropy_chunk, want_entropy=False)
+                                                    tgt, chunk=cfg.entropy_chunk,
+                                                    temperature=cfg.temperature, want_entropy=False)
                 seq_ids = None
                 if cfg.tis_mode == "sequence" or cfg.loss_agg_mode == "seq-mean-token-mean":
+                    # right=True: a loss row landing exactly on a sequence
+                    # boundary (single-token prompt) belongs to the sequence
+                    # that STARTS there, not the previous one
                     seq_ids = torch.bucketize(
-                        rows_idx, torch.tensor(batch.cu_seqlens[1:-1], device=rows_idx.device))
+                        rows_idx, torch.tensor(batch.cu_seqlens[1:-1], device=rows_idx.device),
+                        right=True)
                 tis_w = self._tis_weights(old_lp, rollout_lp, seq_ids=seq_ids)
 
             if micro is micros[0]:
@@ -218,12 +240,13 @@ class PolicyTrainer:
                 loss = loss_tok.sum() / n_global_tokens
             elif cfg.loss_agg_mode == "seq-mean-token-mean":
                 # per-sequence token mean, then mean over the GLOBAL sequence
-                # count (computed once above)
-                loss = 0.0
-                for s in seq_ids.unique():
-                    m = seq_ids == s
-                    loss = loss + loss_tok[m].mean()
-                loss = loss / max(1.0, n_global_rows)
+                # count (computed once above) — one index_add per micro, not
+                # one kernel launch per sequence
+                n_seq = int(seq_ids.max().item()) + 1
+                zeros = torch.zeros(n_seq, dtype=loss_tok.dtype, device=loss_tok.device)
+                seq_sum = zeros.index_add(0, seq_ids, loss_tok)
+                seq_cnt = zeros.index_add(0, seq_ids, torch.ones_like(loss_tok.detach()))
+                loss = (seq_sum / seq_cnt.clamp(min=1.0)).sum() / max(1.0, n_global_rows)
             else:
                 loss = loss_tok.sum() / n_global_tokens
             loss.backward()

@@ -112,8 +112,10 @@ class UnifiedTrainer:
             train_dataset, batch_size=self.config.train_batch_size, seed=self.config.seed,
             rank=pdist.get_rank(), world_size=pdist.get_world_size())
         self.val_dataset = val_dataset
+        # metrics/file backends are rank-0-only under DP (reference logs from
+        # the driver process); other ranks keep a no-op Tracking
         self.tracking = tracking or Tracking(
-            backends=self.config.logger_backends,
+            backends=self.config.logger_backends if pdist.get_rank() == 0 else [],
             project_name=self.config.project_name,
             experiment_name=self.config.experiment_name)
         self.episode_logger = (
@@ -182,10 +184,20 @@ class UnifiedTrainer:
         groups, episodes, rs_metrics = apply_rejection_sampling_and_filtering(
             episodes, groups, self.rs_config, self.rs_state)
         metrics.update(rs_metrics)
-        if not groups:
+        # DP rendezvous on the skip decision: each rank has its own task
+        # shard, so filtering can leave `groups` empty on SOME ranks. A
+        # rank-local early return here would deadlock the others inside
+        # update_policy's collectives (they all-reduce token counts and the
+        # flat grad buffer). Skip only if NO rank has work; otherwise every
+        # rank proceeds — an empty rank contributes zero rows/grads but
+        # still enters every collective.
+        from rllm_amd.parallel import dist as pdist
+
+        n_groups_global = pdist.all_reduce_scalar(float(len(groups)), op="sum")
+        if n_groups_global <= 0:
             metrics["batch/skipped"] = 1.0
             return metrics
-        if self.rs_config.mode == "episode":
+        if groups and self.rs_config.mode == "episode":
             self.rs_state.reset()
 
         # S6 — advantages (before batch build so per-token advs land in rows)
@@ -244,16 +256,23 @@ class UnifiedTrainer:
 
     # ------------------------------------------------------------------
     def save_checkpoint(self):
-        path = Path(self.config.checkpoint_dir) / f"global_step_{self.state.global_step}"
-        path.mkdir(parents=True, exist_ok=True)
-        self.backend.save_checkpoint(str(path), self.state.global_step)
-        import json
+        # Under torchrun DP the replicas hold identical weights after the
+        # all-reduced update — only rank 0 writes, others wait at the
+        # barrier (concurrent writers would corrupt actor.pt / latest.txt).
+        from rllm_amd.parallel import dist as pdist
 
-        (path / "trainer_state.json").write_text(json.dumps(self.state.to_dict()))
-        (path / "data.json").write_text(json.dumps(self.train_loader.state_dict()))
-        latest = Path(self.config.checkpoint_dir) / "latest_checkpointed_iteration.txt"
-        latest.write_text(str(self.state.global_step))
-        logger.info("saved checkpoint at step %d -> %s", self.state.global_step, path)
+        if pdist.get_rank() == 0:
+            path = Path(self.config.checkpoint_dir) / f"global_step_{self.state.global_step}"
+            path.mkdir(parents=True, exist_ok=True)
+            self.backend.save_checkpoint(str(path), self.state.global_step)
+            import json
+
+            (path / "trainer_state.json").write_text(json.dumps(self.state.to_dict()))
+            (path / "data.json").write_text(json.dumps(self.train_loader.state_dict()))
+            latest = Path(self.config.checkpoint_dir) / "latest_checkpointed_iteration.txt"
+            latest.write_text(str(self.state.global_step))
+            logger.info("saved checkpoint at step %d -> %s", self.state.global_step, path)
+        pdist.barrier()
 
     def _try_resume(self):
         import json

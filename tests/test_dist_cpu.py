@@ -232,3 +232,82 @@ def _worker_default_group(rank, world, q):
 @pytest.mark.timeout(180)
 def test_default_group_scoping_world2():
     _run_spawn(_worker_default_group, "29619")
+
+
+def _worker_empty_rank_rendezvous(rank, world, q):
+    """Deadlock guard: rank 1's only group is filtered out
+    (uniform rewards) while rank 0 keeps one — BOTH ranks must still enter
+    update_policy's collectives, the empty rank with zero rows."""
+    try:
+        import asyncio
+
+        pdist = _setup(rank, world)
+
+        from rllm_amd.trainer.algorithms.config import RejectionSamplingConfig
+        from rllm_amd.trainer.backend_protocol import BackendProtocol
+        from rllm_amd.trainer.batch import rows_from_groups
+        from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+        from rllm_amd.types import Episode, Step, Trajectory
+
+        class FakeBackend(BackendProtocol):
+            def __init__(self, rank):
+                self.rank = rank
+                self.update_calls = []
+
+            def init_rollout_engine(self):
+                return None
+
+            async def generate_episodes(self, tasks, uids=None, is_validation=False):
+                eps = []
+                for i, uid in enumerate(uids):
+                    # rank 1: uniform rewards -> group filtered out
+                    r = 1.0 if self.rank == 1 else float(i % 2)
+                    step = Step(prompt_ids=[1, 2, 3], response_ids=[4, 5],
+                                logprobs=[-0.1, -0.2],
+                                chat_completions=[{"role": "user", "content": "x"}],
+                                reward=r, done=True)
+                    eps.append(Episode(id=uid, task={"q": "x"},
+                                       trajectories=[Trajectory(name="solver", steps=[step], reward=r)],
+                                       is_correct=r > 0))
+                return eps
+
+            def transform_to_backend_batch(self, groups):
+                return rows_from_groups(groups)
+
+            def update_policy(self, rows):
+                # mimic PolicyTrainer's collective pattern: every rank MUST
+                # reach these or the others hang
+                n = float(sum(sum(r.response_mask) for r in rows))
+                n_global = pdist.all_reduce_scalar(n, op="sum")
+                t = torch.zeros(4)
+                pdist.all_reduce_sum_(t)
+                self.update_calls.append((len(rows), n_global))
+                return {"actor/n_rows": float(len(rows))}
+
+        backend = FakeBackend(rank)
+        tasks = [{"q": f"t{i}"} for i in range(2)]  # 1 task per rank
+        trainer = UnifiedTrainer(
+            backend, tasks,
+            config=TrainerConfig(train_batch_size=1, rollout_n=2, max_steps=1,
+                                 logger_backends=[]),
+            rejection_sampling_config=RejectionSamplingConfig(
+                mode="none", filter_uniform_groups=True))
+        trainer.fit()
+
+        # both ranks made exactly one update_policy call and agree globally
+        assert len(backend.update_calls) == 1, backend.update_calls
+        n_rows, n_global = backend.update_calls[0]
+        assert n_global > 0
+        if rank == 1:
+            assert n_rows == 0  # filtered rank participated with an empty batch
+        else:
+            assert n_rows == 2
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+@pytest.mark.timeout(180)
+def test_empty_rank_rendezvous_world2():
+    _run_spawn(_worker_empty_rank_rendezvous, "29621")
