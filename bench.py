@@ -52,6 +52,9 @@ def parse_args():
                    choices=["recompute", "alias", "rollout"],
                    help="'alias' is bit-identical with one optimizer step per batch but skips a forward")
     p.add_argument("--seed", type=int, default=1234)
+    p.add_argument("--dist-backend", default=None, choices=[None, "nccl", "gloo"],
+                   help="override the collective backend (gloo lets a 2-rank "
+                        "smoke run share ONE GPU, which RCCL forbids)")
     return p.parse_args()
 
 
@@ -75,8 +78,11 @@ def main():
     from rllm_amd.trainer.policy import PolicyTrainer, PolicyTrainerConfig
     from rllm_amd.types import Episode, Step, Trajectory
 
-    pdist.init_from_env()
-    device = f"cuda:{local_rank}"
+    pdist.init_from_env(backend=args.dist_backend)
+    # gloo smoke mode: ranks share GPU 0 (a 1-GPU box); RCCL mode: one GPU
+    # per rank as torchrun assigns
+    n_devices = torch.cuda.device_count()
+    device = f"cuda:{local_rank % n_devices}"
     torch.cuda.set_device(device)
 
     cfg = get_model_config(args.model)

