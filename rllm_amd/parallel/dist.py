@@ -34,7 +34,8 @@ def init_from_env(backend: str | None = None, timeout_s: int = 1800) -> tuple[in
         dist.init_process_group(backend=backend, rank=rank, world_size=world_size,
                                 timeout=timedelta(seconds=timeout_s))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
+        # modulo: a world-2 gloo smoke on a 1-GPU box shares device 0
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
     return rank, world_size, local_rank
 
 

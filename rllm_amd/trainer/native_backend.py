@@ -65,9 +65,27 @@ class NativeBackend(BackendProtocol):
     ):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
-        self.cfg = get_model_config(model_config) if isinstance(model_config, str) else model_config
+        # model_config may be a registry name, a ModelConfig, or a LOCAL
+        # HuggingFace model dir (config.json + *.safetensors) — the real-
+        # model path (reference train_agent_ppo.py:108-126 tokenizer +
+        # weight load)
+        self.hf_model_dir: str | None = None
+        if isinstance(model_config, str):
+            from rllm_amd.models.hf_loader import config_from_hf, is_hf_model_dir
+
+            if is_hf_model_dir(model_config):
+                self.hf_model_dir = model_config
+                self.cfg = config_from_hf(model_config)
+            else:
+                self.cfg = get_model_config(model_config)
+        else:
+            self.cfg = model_config
         self.device = device
         self.seed = seed
+        if tokenizer is None and self.hf_model_dir is not None:
+            from rllm_amd.utils.tokenizer import load_tokenizer
+
+            tokenizer = load_tokenizer(self.hf_model_dir)
         self.tokenizer = tokenizer or ByteTokenizer()
         self.parser = parser or QwenChatTemplateParser(self.tokenizer)
         self.policy_config = policy_config or PolicyTrainerConfig()
@@ -109,7 +127,12 @@ class NativeBackend(BackendProtocol):
         if self.device == "cuda" and "LOCAL_RANK" in os.environ:
             self.device = f"cuda:{os.environ['LOCAL_RANK']}"
             torch.cuda.set_device(self.device)
-        self.model = QwenModel(self.cfg, device=self.device).init_random(seed=self.seed)
+        if self.hf_model_dir is not None:
+            from rllm_amd.models.hf_loader import load_hf_model
+
+            self.model, _ = load_hf_model(self.hf_model_dir, device=self.device)
+        else:
+            self.model = QwenModel(self.cfg, device=self.device).init_random(seed=self.seed)
         if self.checkpoint_path:
             sd = torch.load(self.checkpoint_path, weights_only=True, map_location=self.device)
             self.model.load_state_dict(sd)

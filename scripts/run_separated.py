@@ -81,7 +81,8 @@ def main():
     from rllm_amd.trainer.policy import PolicyTrainer, PolicyTrainerConfig
 
     trainer = PolicyTrainer(model, None, PolicyTrainerConfig(lr=1e-6, kl_beta=0.0,
-                                                             use_ref=False))
+                                                             use_ref=False,
+                                                             old_logprob_mode="rollout"))
     client = SeparatedRolloutClient(topo, trainer.flat_param)
     rng = np.random.default_rng(args.seed)
 

@@ -147,7 +147,8 @@ def test_update_policy_end_to_end():
     torch.manual_seed(4)
     model = tiny_model(seed=21)
     ref = tiny_model(seed=21)
-    trainer = PolicyTrainer(model, ref, PolicyTrainerConfig(lr=1e-3, kl_beta=0.0, grad_clip=10.0))
+    trainer = PolicyTrainer(model, ref, PolicyTrainerConfig(
+        lr=1e-3, kl_beta=0.0, grad_clip=10.0, old_logprob_mode="rollout"))
 
     engine = LLMEngine(model, kv_budget_bytes=32 << 20, eos_token_id=None, seed=9)
     prompts = [list(range(5, 37)) for _ in range(8)]

@@ -435,7 +435,7 @@ def test_update_policy_seq_mean_mode():
     model = QwenModel(cfg, device="cuda").init_random(seed=3)
     pt = PolicyTrainer(model, None, PolicyTrainerConfig(
         lr=1e-4, kl_beta=0.0, use_ref=False, loss_agg_mode="seq-mean-token-mean",
-        max_tokens_per_micro=64))  # forces MULTIPLE micros
+        old_logprob_mode="rollout", max_tokens_per_micro=64))  # forces MULTIPLE micros
     rng = random.Random(0)
     rows = []
     for _ in range(6):
