@@ -48,9 +48,12 @@ def parse_args():
                    help="ppo_max_token_len_per_gpu (0 = auto by model size)")
     p.add_argument("--kl-beta", type=float, default=1e-3)
     p.add_argument("--lr", type=float, default=1e-6)
-    p.add_argument("--old-logprob-mode", default="recompute",
+    p.add_argument("--old-logprob-mode", default="alias",
                    choices=["recompute", "alias", "rollout"],
-                   help="'alias' is bit-identical with one optimizer step per batch but skips a forward")
+                   help="'alias' (default) is BIT-identical to 'recompute' at one optimizer "
+                        "step per batch (proven by tests/test_engine_gpu.py::"
+                        "test_alias_old_logprob_bit_identical_to_recompute) but skips the "
+                        "redundant no-grad forward")
     p.add_argument("--seed", type=int, default=1234)
     p.add_argument("--dist-backend", default=None, choices=[None, "nccl", "gloo"],
                    help="override the collective backend (gloo lets a 2-rank "
@@ -212,6 +215,10 @@ def main():
                 "temperature": 1.0, "top_p": 1.0, "clip_ratio": 0.2,
                 "kl_beta": args.kl_beta, "lr": args.lr,
                 "loss_agg": "token-mean",
+                "old_logprob_mode": args.old_logprob_mode + (
+                    " (bit-identical to recompute at 1 optim step/batch; "
+                    "test_alias_old_logprob_bit_identical_to_recompute)"
+                    if args.old_logprob_mode == "alias" else ""),
                 "rollout_ms_per_step": round(phase_times["rollout_s"] / args.steps * 1000, 1),
                 "update_ms_per_step": round(phase_times["update_s"] / args.steps * 1000, 1),
                 "note": "tokens = response tokens generated AND trained on per wall-clock second, whole job",

@@ -153,6 +153,11 @@ class LLMEngine:
         self._graph_pool = None
         self._gb = None                # static device buffers dict
         self._hb = None                # pinned host staging dict
+        # ---- hipGraph prefill state: the training workloads prefill in
+        # UNIFORM chunks (max_num_batched_tokens), so a handful of exact
+        # (T, n_tiles) keys cover every call; anything else runs eager ----
+        self._prefill_graphs: dict = {}  # (T, n_tiles) -> dict(graph, dev, host, hidden)
+        self._prefill_graph_cap = 8
 
     # ------------------------------------------------------------------
     @property
@@ -311,15 +316,88 @@ class LLMEngine:
             row += n
             last_rows.append(row - 1)
 
-        ids_t = torch.tensor(input_ids, device=device, dtype=torch.long)
-        pos_t = torch.tensor(positions, device=device, dtype=torch.int32)
-        slots_t = torch.tensor(slot_mapping, device=device, dtype=torch.int32)
-        tiles = make_prefill_tiles(seqlens, device)
-
-        hidden = self.model.forward_prefill(ids_t, pos_t, tiles, self.kv, slots_t)
+        T = len(input_ids)
+        tile = 64
+        n_tiles = sum((n + tile - 1) // tile for n in seqlens)
+        key = (T, n_tiles)
+        if self.use_hip_graph and (key in self._prefill_graphs
+                                   or len(self._prefill_graphs) < self._prefill_graph_cap):
+            hidden = self._run_prefill_graph(key, input_ids, positions, slot_mapping, seqlens)
+        else:
+            ids_t = torch.tensor(input_ids, device=device, dtype=torch.long)
+            pos_t = torch.tensor(positions, device=device, dtype=torch.int32)
+            slots_t = torch.tensor(slot_mapping, device=device, dtype=torch.int32)
+            tiles = make_prefill_tiles(seqlens, device)
+            hidden = self.model.forward_prefill(ids_t, pos_t, tiles, self.kv, slots_t)
         last_hidden = hidden[torch.tensor(last_rows, device=device, dtype=torch.long)]
         self._sample_and_append(batch, last_hidden)
         return len(input_ids)
+
+    def _run_prefill_graph(self, key: tuple, input_ids, positions, slot_mapping, seqlens):
+        """Replay (or capture) the full-prefill forward for this exact
+        (token count, tile count) shape. Inputs stage through pinned host
+        buffers into the captured static device buffers; sampling and the
+        last-row gather stay eager (per-seq params). ~250 launches/call
+        collapse into one replay (TODO r1 #6: ~0.4 s/step of prefill launch
+        overhead at the flagship config)."""
+        T, n_tiles = key
+        dev = self.device
+        entry = self._prefill_graphs.get(key)
+        if entry is None:
+            dbuf = {
+                "ids": torch.zeros(T, device=dev, dtype=torch.long),
+                "pos": torch.zeros(T, device=dev, dtype=torch.int32),
+                "slot": torch.zeros(T, device=dev, dtype=torch.int32),
+                "tss": torch.zeros(n_tiles, device=dev, dtype=torch.int32),
+                "tr0": torch.zeros(n_tiles, device=dev, dtype=torch.int32),
+                "tsl": torch.zeros(n_tiles, device=dev, dtype=torch.int32),
+            }
+            hbuf = {k: torch.zeros(v.shape, dtype=v.dtype, pin_memory=True)
+                    for k, v in dbuf.items()}
+            entry = {"dev": dbuf, "host": hbuf, "graph": None, "hidden": None}
+            self._prefill_graphs[key] = entry
+        dbuf, hbuf = entry["dev"], entry["host"]
+
+        # stage host → pinned → device
+        hbuf["ids"].numpy()[:] = input_ids
+        hbuf["pos"].numpy()[:] = positions
+        hbuf["slot"].numpy()[:] = slot_mapping
+        tss, tr0, tsl = [], [], []
+        start = 0
+        for n in seqlens:
+            for r0 in range(0, n, 64):
+                tss.append(start)
+                tr0.append(start + r0)
+                tsl.append(n)
+            start += n
+        hbuf["tss"].numpy()[:] = tss
+        hbuf["tr0"].numpy()[:] = tr0
+        hbuf["tsl"].numpy()[:] = tsl
+        for k in dbuf:
+            dbuf[k].copy_(hbuf[k], non_blocking=True)
+
+        def compute():
+            return self.model.forward_prefill(
+                dbuf["ids"], dbuf["pos"], (dbuf["tss"], dbuf["tr0"], dbuf["tsl"]),
+                self.kv, dbuf["slot"])
+
+        if entry["graph"] is None:
+            # warm up eager once (workspace/algo init), then capture
+            torch.cuda.synchronize()
+            compute()
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            if self._graph_pool is None:
+                with torch.cuda.graph(graph):
+                    hidden = compute()
+                self._graph_pool = graph.pool()
+            else:
+                with torch.cuda.graph(graph, pool=self._graph_pool):
+                    hidden = compute()
+            entry["graph"], entry["hidden"] = graph, hidden
+        # capture does NOT execute the kernels — always replay
+        entry["graph"].replay()
+        return entry["hidden"]
 
     def _run_prefill_cached(self, batch: list[Sequence]) -> int:
         """Suffix-only prefill: cached prefix K/V come from shared pages;

@@ -57,6 +57,9 @@ class PolicyTrainerConfig:
     # meta_info['temperature'], verl_backend.py:612).
     temperature: float = 1.0
     entropy_chunk: int = 16384
+    # run the frozen-ref no-grad forward of micro i+1 on a second HIP
+    # stream, overlapped with micro i's backward (single-GPU latency hide)
+    overlap_ref_stream: bool = True
     use_ref: bool = True
     # LoRA training: the KL reference is the actor's own base weights with
     # adapters disabled (models/lora.py) — no second model copy.
@@ -172,8 +175,42 @@ class PolicyTrainer:
         tot_clip = 0.0
         ent_sum, ent_n = 0.0, 0
 
-        for micro in micros:
-            batch = pack_rows(micro, device=str(self.device))
+        # pack every micro up front so the ref-prefetch stream can read
+        # batch i+1's tensors while micro i is still in backward
+        packed = [pack_rows(m, device=str(self.device)) for m in micros]
+        use_ref_any = cfg.use_ref and (cfg.ref_from_lora_base or self.ref_model is not None)
+        # overlap the (no-grad) frozen-reference forward of micro i+1 with
+        # micro i's backward on a second HIP stream: the ref path is pure
+        # torch.matmul + stateless HIP kernels (no shared workspace), so the
+        # streams don't contend on state — only on CUs
+        overlap = (cfg.overlap_ref_stream and use_ref_any and len(packed) > 1
+                   and getattr(self.device, "type", str(self.device)) == "cuda")
+        ref_stream = torch.cuda.Stream(device=self.device) if overlap else None
+        if overlap:
+            ref_stream.wait_stream(torch.cuda.current_stream())  # packs visible
+        prefetched: dict[int, torch.Tensor] = {}
+
+        @torch.no_grad()
+        def ref_lp_of(i: int) -> torch.Tensor:
+            b = packed[i]
+            ridx = b.loss_mask.nonzero(as_tuple=True)[0]
+            if cfg.ref_from_lora_base:
+                from rllm_amd.models import lora as _lora
+
+                with _lora.disabled(self.model):
+                    rh = self.model.forward_train(b.input_ids, b.positions, b.cu_seqlens)
+                    rlp, _ = ops.chunked_logprob(rh[ridx], self.model.lm_weight,
+                                                 b.targets[ridx], chunk=cfg.entropy_chunk,
+                                                 temperature=cfg.temperature, want_entropy=False)
+            else:
+                rh = self.ref_model.forward_train(b.input_ids, b.positions, b.cu_seqlens)
+                rlp, _ = ops.chunked_logprob(rh[ridx], self.ref_model.lm_weight,
+                                             b.targets[ridx], chunk=cfg.entropy_chunk,
+                                             temperature=cfg.temperature, want_entropy=False)
+            return rlp
+
+        for mi, batch in enumerate(packed):
+            micro = micros[mi]
             hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
             rows_idx = batch.loss_mask.nonzero(as_tuple=True)[0]
             h = hidden[rows_idx]
@@ -203,19 +240,13 @@ class PolicyTrainer:
                 else:
                     old_lp = old_logprob_fn(batch, rows_idx)
                 ref_lp = None
-                if cfg.use_ref and cfg.ref_from_lora_base:
-                    from rllm_amd.models import lora as _lora
-
-                    with _lora.disabled(self.model):
-                        ref_hidden = self.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
-                        ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.model.lm_weight,
-                                                        tgt, chunk=cfg.entropy_chunk,
-                                                        temperature=cfg.temperature, want_entropy=False)
-                elif cfg.use_ref and self.ref_model is not None:
-                    ref_hidden = self.ref_model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
-                    ref_lp, _ = ops.chunked_logprob(ref_hidden[rows_idx], self.ref_model.lm_weight,
-                                                    tgt, chunk=cfg.entropy_chunk,
-                                                    temperature=cfg.temperature, want_entropy=False)
+                if use_ref_any:
+                    if mi in prefetched:
+                        torch.cuda.current_stream().wait_stream(ref_stream)
+                        ref_lp = prefetched.pop(mi)
+                        ref_lp.record_stream(torch.cuda.current_stream())
+                    else:
+                        ref_lp = ref_lp_of(mi)
                 seq_ids = None
                 if cfg.tis_mode == "sequence" or cfg.loss_agg_mode == "seq-mean-token-mean":
                     # right=True: a loss row landing exactly on a sequence
@@ -226,7 +257,7 @@ class PolicyTrainer:
                         right=True)
                 tis_w = self._tis_weights(old_lp, rollout_lp, seq_ids=seq_ids)
 
-            if micro is micros[0]:
+            if mi == 0:
                 from rllm_amd.utils.offpolicy import compute_offpolicy_metrics
 
                 offpolicy_metrics = compute_offpolicy_metrics(old_lp, rollout_lp)
@@ -250,6 +281,11 @@ class PolicyTrainer:
             else:
                 loss = loss_tok.sum() / n_global_tokens
             loss.backward()
+            # prefetch the NEXT micro's frozen-ref forward concurrently with
+            # this micro's backward
+            if overlap and mi + 1 < len(packed) and (mi + 1) not in prefetched:
+                with torch.cuda.stream(ref_stream):
+                    prefetched[mi + 1] = ref_lp_of(mi + 1)
             tot_loss += float(loss.detach()) * (1.0 if cfg.loss_agg_mode != "token-mean" else 1.0)
             tot_pg += float(loss_tok.detach().sum())
             tot_clip += float(clipped.sum())

@@ -244,3 +244,56 @@ def test_max_model_len_enforced():
     assert outs["capped"].finish_reason == "length"
     # stopped at the window: prompt 40 + completion <= 48 (+1 step slack)
     assert len(outs["capped"].token_ids) <= 9
+
+
+@requires_gpu
+def test_alias_old_logprob_bit_identical_to_recompute():
+    """old_logprob_mode='alias' (old_lp := lp.detach()) must be BIT-identical
+    to 'recompute' (a second no-grad forward with the same pre-step weights):
+    same deterministic kernels, same weights, one optimizer step per
+    update_policy call. This is the exactness test gating the bench default
+    (VERDICT r1 #4; +11% measured from skipping the recompute forward)."""
+    import random
+
+    from rllm_amd.trainer.batch import PackedRow
+
+    def build(mode):
+        torch.manual_seed(4)
+        model = tiny_model(seed=33)
+        ref = tiny_model(seed=33)
+        for p in ref.parameters():
+            p.requires_grad_(False)
+        return PolicyTrainer(model, ref, PolicyTrainerConfig(
+            lr=1e-3, kl_beta=1e-2, grad_clip=1.0, old_logprob_mode=mode,
+            max_tokens_per_micro=128))  # multiple micros
+
+    rng = random.Random(7)
+    rows = []
+    for _ in range(6):
+        n = rng.randint(24, 48)
+        p = rng.randint(4, 8)
+        rows.append(PackedRow(tokens=[rng.randrange(1024) for _ in range(n)],
+                              response_mask=[0] * p + [1] * (n - p),
+                              advantages=[rng.uniform(-1, 1)] * n,
+                              rollout_logprobs=[-1.0] * n))
+
+    t_re = build("recompute")
+
+    def old_lp_fn(batch, rows_idx):
+        with torch.no_grad():
+            h = t_re.model.forward_train(batch.input_ids, batch.positions, batch.cu_seqlens)
+            lp, _ = ops.chunked_logprob(h[rows_idx], t_re.model.lm_weight,
+                                        batch.targets[rows_idx], want_entropy=False)
+            return lp
+
+    m_re = t_re.update_policy(rows, old_logprob_fn=old_lp_fn)
+    t_al = build("alias")
+    m_al = t_al.update_policy(rows)
+
+    # bit-identical weights after the step
+    assert torch.equal(t_re.flat_param, t_al.flat_param), \
+        (t_re.flat_param - t_al.flat_param).abs().max()
+    for k in ("actor/loss", "actor/grad_norm", "actor/clipfrac"):
+        assert m_re[k] == m_al[k], (k, m_re[k], m_al[k])
+    # and the off-policy diagnostics agree old_lp == recomputed lp exactly
+    assert m_al["offpolicy/abs_diff_max"] == 0.0
