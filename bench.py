@@ -114,9 +114,11 @@ def main():
         ))
 
     free, _ = torch.cuda.mem_get_info()
+    # ranks sharing one device (gloo smoke on a 1-GPU box) split the budget
+    ranks_per_dev = max(1, (world + n_devices - 1) // n_devices)
     engine = LLMEngine(
         model, max_num_seqs=1024, max_num_batched_tokens=8192,
-        kv_budget_bytes=min(int(free * 0.35), 64 << 30),
+        kv_budget_bytes=min(int(free * 0.35) // ranks_per_dev, 64 << 30),
         eos_token_id=None,  # synthetic data: length-capped rollouts
         seed=args.seed * 1000 + rank)
 

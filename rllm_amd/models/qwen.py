@@ -224,13 +224,33 @@ class QwenLayer(nn.Module):
         cfg = self.cfg
         T = h.shape[0]
         k_pages, v_pages = kv_cache[layer_idx]
-        x = ops.add_rmsnorm_(h, delta, self.input_layernorm, cfg.rms_eps)
-        qkv = self._lin_decode(x, "qkv_proj")
+        fp8 = getattr(self, "_fp8", None)
+        # fused-quant fp8 path (TODO r1 #12): the producing kernels emit
+        # e4m3 directly for the qkv/gate_up/down GEMM inputs — no standalone
+        # quant launches. o_proj keeps the delayed-quant inside _lin_decode
+        # (its input comes from paged_decode's merge).
+        fuse8 = (fp8 is not None and not self._lora_on
+                 and all(n in fp8 for n in ("qkv_proj", "gate_up_proj", "down_proj")))
+        if fuse8:
+            w8q, swq, sq, aq = fp8["qkv_proj"]
+            x8 = ops.add_rmsnorm_fp8_(h, delta, self.input_layernorm, cfg.rms_eps, sq, aq)
+            qkv = ops.fp8_mm_prequant(x8, w8q, sq, swq)
+        else:
+            x = ops.add_rmsnorm_(h, delta, self.input_layernorm, cfg.rms_eps)
+            qkv = self._lin_decode(x, "qkv_proj")
         q = ops.qkv_rope_cache(qkv, self.qkv_bias, k_pages, v_pages, cos_t, sin_t,
                                positions, slot_mapping, cfg.num_heads, cfg.num_kv_heads)
         attn = ops.paged_decode(q, k_pages, v_pages, block_tables, seq_lens,
                                 1.0 / math.sqrt(cfg.head_dim))
         attn_delta = self._tp_reduce(self._lin_decode(attn.reshape(T, cfg.q_size), "o_proj"))
+        if fuse8:
+            w8g, swg, sg, ag = fp8["gate_up_proj"]
+            x8 = ops.add_rmsnorm_fp8_(h, attn_delta, self.post_attention_layernorm,
+                                      cfg.rms_eps, sg, ag)
+            gu = ops.fp8_mm_prequant(x8, w8g, sg, swg)
+            w8d, swd, sd, ad = fp8["down_proj"]
+            m8 = ops.swiglu_fp8(gu, sd, ad)
+            return self._tp_reduce(ops.fp8_mm_prequant(m8, w8d, sd, swd))
         x = ops.add_rmsnorm_(h, attn_delta, self.post_attention_layernorm, cfg.rms_eps)
         mlp_delta = self._tp_reduce(
             self._lin_decode(ops.swiglu(self._lin_decode(x, "gate_up_proj")), "down_proj"))

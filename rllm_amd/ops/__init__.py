@@ -360,6 +360,24 @@ def fp8_linear_delayed(x: torch.Tensor, w8: torch.Tensor, sw: torch.Tensor,
     return y
 
 
+def add_rmsnorm_fp8_(h, delta, weight, eps, scale, amax):
+    """Fused residual add (in place on h) + RMSNorm emitting e4m3 with
+    delayed scaling — the producing kernel quantizes the next GEMM's input
+    directly (no separate quant launch, no bf16 round trip)."""
+    return require_ext().add_rmsnorm_fp8_(h, delta, weight, eps, scale, amax)
+
+
+def swiglu_fp8(gateup, scale, amax):
+    """SwiGLU emitting e4m3 with delayed scaling (input to the down GEMM)."""
+    return require_ext().swiglu_fp8(gateup, scale, amax)
+
+
+def fp8_mm_prequant(x8, w8, sx, sw):
+    """Tuned fp8 GEMM over a PRE-quantized activation (produced by
+    add_rmsnorm_fp8_/swiglu_fp8); bf16 out."""
+    return require_ext().hbl_fp8_mm(x8, w8, sx, sw)
+
+
 def fp8_scale_update_all(scales: torch.Tensor, amaxes: torch.Tensor):
     """Fold every call site's accumulated amax into its scale (one launch;
     called at the top of each decode step)."""

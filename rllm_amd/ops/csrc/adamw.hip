@@ -40,7 +40,22 @@ __global__ void grad_sq_sum_kernel(
     acc += g * g;
   }
   acc = block_reduce_sum(acc, scratch);
-  if (threadIdx.x == 0) atomicAdd(out, acc);
+  // DETERMINISTIC reduction: per-block partial to its own slot; a fixed-
+  // order second pass sums them. (atomicAdd here made the grad norm — and
+  // through grad clipping the whole optimizer step — vary by ~1 ulp run to
+  // run, breaking bit-reproducibility.)
+  if (threadIdx.x == 0) out[blockIdx.x] = acc;
+}
+
+__global__ void sum_partials_kernel(const float* __restrict__ partials,
+                                    float* __restrict__ out, int n) {
+  // one block, fixed order: lane-strided partial sums then an ORDERED
+  // tree over the 4 wave results via LDS
+  __shared__ float scratch[16];
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) acc += partials[i];
+  acc = block_reduce_sum(acc, scratch);
+  if (threadIdx.x == 0) out[0] = acc;
 }
 
 __global__ void adamw_kernel(
@@ -81,10 +96,16 @@ static inline hipStream_t aw_stream() {
 
 torch::Tensor grad_sq_sum(torch::Tensor grad, double grad_scale) {
   TORCH_CHECK(grad.is_cuda() && grad.dtype() == torch::kBFloat16 && grad.is_contiguous());
-  auto out = torch::zeros({1}, grad.options().dtype(torch::kFloat32));
   const int64_t N = grad.numel();
-  hipLaunchKernelGGL(grad_sq_sum_kernel, dim3(grid_for(N / 8, 256)), dim3(256), 0, aw_stream(),
-                     (const uint16_t*)grad.data_ptr(), out.data_ptr<float>(), N, (float)grad_scale);
+  const unsigned grid = grid_for(N / 8, 256);
+  auto partials = torch::empty({(int64_t)grid}, grad.options().dtype(torch::kFloat32));
+  auto out = torch::zeros({1}, grad.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(grad_sq_sum_kernel, dim3(grid), dim3(256), 0, aw_stream(),
+                     (const uint16_t*)grad.data_ptr(), partials.data_ptr<float>(), N,
+                     (float)grad_scale);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(sum_partials_kernel, dim3(1), dim3(256), 0, aw_stream(),
+                     partials.data_ptr<float>(), out.data_ptr<float>(), (int)grid);
   HIP_CHECK_KERNEL();
   return out;
 }

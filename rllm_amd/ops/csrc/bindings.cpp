@@ -70,6 +70,10 @@ torch::Tensor hbl_fp8_mm(torch::Tensor x8, torch::Tensor w8, torch::Tensor sx, t
 // elementwise.hip (fp8)
 void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale, torch::Tensor amax_next);
 void fp8_scale_update(torch::Tensor scale, torch::Tensor amax_next);
+torch::Tensor add_rmsnorm_fp8_(torch::Tensor h, c10::optional<torch::Tensor> delta,
+                               torch::Tensor w, double eps,
+                               torch::Tensor scale, torch::Tensor amax_next);
+torch::Tensor swiglu_fp8(torch::Tensor gateup, torch::Tensor scale, torch::Tensor amax_next);
 
 // sampling.hip
 std::vector<torch::Tensor> sample_logprob(torch::Tensor logits, double temperature,
@@ -106,6 +110,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hbl_fp8_mm", &hbl_fp8_mm, "Tuned fp8 GEMM with device scale pointers");
   m.def("fp8_quant_delayed", &fp8_quant_delayed, "Fused delayed-scaling e4m3 quantization");
   m.def("fp8_scale_update", &fp8_scale_update, "Fold accumulated amax into the fp8 scale");
+  m.def("add_rmsnorm_fp8_", &add_rmsnorm_fp8_,
+        "Fused residual add + RMSNorm emitting e4m3 (delayed scaling)");
+  m.def("swiglu_fp8", &swiglu_fp8, "SwiGLU emitting e4m3 (delayed scaling)");
   m.def("sample_logprob", &sample_logprob, "Fused gumbel-max sampling + logprob");
   m.def("gather_logprob", &gather_logprob, "Logprob of given tokens from logits");
 }

@@ -549,6 +549,165 @@ void fp8_quant_delayed(torch::Tensor x, torch::Tensor y, torch::Tensor scale,
   HIP_CHECK_KERNEL();
 }
 
+// ---------------------------------------------------------------------------
+// FUSED fp8 producers (round-2, TODO #12): the kernel that PRODUCES a GEMM
+// activation emits e4m3 directly (same delayed-scaling recipe as
+// fp8_quant_delayed) — erases one full read+write of the activation stream
+// and one kernel launch per GEMM site. Used on the decode hot path for the
+// qkv / gate_up / down inputs.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void store_fp8_8(uint8_t* dst, const float* v,
+                                            float inv_s, float& local_max) {
+  auto cl = [](float x) { return fminf(fmaxf(x, -448.f), 448.f); };
+  int p0 = 0, p1 = 0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) local_max = fmaxf(local_max, fabsf(v[j]));
+  p0 = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v[0] * inv_s), cl(v[1] * inv_s), p0, false);
+  p0 = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v[2] * inv_s), cl(v[3] * inv_s), p0, true);
+  p1 = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v[4] * inv_s), cl(v[5] * inv_s), p1, false);
+  p1 = __builtin_amdgcn_cvt_pk_fp8_f32(cl(v[6] * inv_s), cl(v[7] * inv_s), p1, true);
+  uint2 packed = make_uint2((uint32_t)p0, (uint32_t)p1);
+  *reinterpret_cast<uint2*>(dst) = packed;
+}
+
+// add_rmsnorm with e4m3 output: h += delta (bf16, in place); y8 = e4m3 of
+// rmsnorm(h)*w. One wave per row (H % 8 == 0).
+__global__ void add_rmsnorm_fp8_wave_kernel(
+    uint16_t* __restrict__ h,
+    const uint16_t* __restrict__ delta,
+    const uint16_t* __restrict__ w,
+    uint8_t* __restrict__ y8,
+    const float* __restrict__ scale,
+    float* __restrict__ amax_next,
+    int64_t T, int H, float eps) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t row = (int64_t)blockIdx.x * 4 + wid;
+  if (row >= T) return;
+  uint16_t* hr = h + row * (int64_t)H;
+  const uint16_t* dr = delta ? delta + row * (int64_t)H : nullptr;
+  uint8_t* yr = y8 + row * (int64_t)H;
+  const float inv_s = 1.f / scale[0];
+
+  float ss = 0.f;
+  for (int i = lane * 8; i < H; i += WAVE_SIZE * 8) {
+    short8v a = *reinterpret_cast<const short8v*>(hr + i);
+    if (dr) {
+      short8v da = *reinterpret_cast<const short8v*>(dr + i);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = bf16_to_f32((uint16_t)a[j]) + bf16_to_f32((uint16_t)da[j]);
+        o[j] = (short)f32_to_bf16(v);
+        ss += v * v;
+      }
+      *reinterpret_cast<short8v*>(hr + i) = o;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = bf16_to_f32((uint16_t)a[j]);
+        ss += v * v;
+      }
+    }
+  }
+#pragma unroll
+  for (int off = 1; off < WAVE_SIZE; off <<= 1) ss += __shfl_xor(ss, off, WAVE_SIZE);
+  const float inv_rms = rsqrtf(ss / (float)H + eps);
+  float local_max = 0.f;
+  for (int i = lane * 8; i < H; i += WAVE_SIZE * 8) {
+    short8v a = *reinterpret_cast<const short8v*>(hr + i);
+    short8v wv = *reinterpret_cast<const short8v*>(w + i);
+    float v[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = bf16_to_f32((uint16_t)a[j]) * inv_rms * bf16_to_f32((uint16_t)wv[j]);
+    store_fp8_8(yr + i, v, inv_s, local_max);
+  }
+#pragma unroll
+  for (int off = 1; off < WAVE_SIZE; off <<= 1)
+    local_max = fmaxf(local_max, __shfl_xor(local_max, off, WAVE_SIZE));
+  if (lane == 0 && local_max > 0.f)
+    atomicMax(reinterpret_cast<unsigned int*>(amax_next), __float_as_uint(local_max));
+}
+
+torch::Tensor add_rmsnorm_fp8_(torch::Tensor h, c10::optional<torch::Tensor> delta,
+                               torch::Tensor w, double eps,
+                               torch::Tensor scale, torch::Tensor amax_next) {
+  TORCH_CHECK(h.is_cuda() && h.dtype() == torch::kBFloat16 && h.is_contiguous());
+  const int64_t T = h.numel() / h.size(-1);
+  const int H = (int)h.size(-1);
+  TORCH_CHECK(H % 8 == 0, "add_rmsnorm_fp8_ needs H % 8 == 0");
+  auto y = torch::empty(h.sizes(), h.options().dtype(torch::kFloat8_e4m3fn));
+  const uint16_t* dptr = nullptr;
+  if (delta.has_value()) {
+    TORCH_CHECK(delta->is_contiguous() && delta->sizes() == h.sizes());
+    dptr = (const uint16_t*)delta->data_ptr();
+  }
+  hipLaunchKernelGGL(add_rmsnorm_fp8_wave_kernel, dim3((unsigned)((T + 3) / 4)), dim3(256),
+                     0, cur_stream(), (uint16_t*)h.data_ptr(), dptr,
+                     (const uint16_t*)w.data_ptr(), (uint8_t*)y.data_ptr(),
+                     scale.data_ptr<float>(), amax_next.data_ptr<float>(), T, H, (float)eps);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+// SwiGLU with e4m3 output (gateup [T, 2I] bf16 -> y8 [T, I] e4m3)
+__global__ void swiglu_fp8_kernel8(
+    const uint16_t* __restrict__ gateup,
+    uint8_t* __restrict__ out,
+    const float* __restrict__ scale,
+    float* __restrict__ amax_next,
+    int64_t T, int I) {
+  __shared__ float red[16];
+  const float inv_s = 1.f / scale[0];
+  float local_max = 0.f;
+  const int64_t total = T * (int64_t)I / 8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = idx / (I / 8);
+    int i = (int)(idx % (I / 8)) * 8;
+    short8v gv = *reinterpret_cast<const short8v*>(gateup + t * 2 * I + i);
+    short8v uv = *reinterpret_cast<const short8v*>(gateup + t * 2 * I + I + i);
+    float v[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32((uint16_t)gv[j]);
+      float uf = bf16_to_f32((uint16_t)uv[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      v[j] = gf * sig * uf;
+    }
+    store_fp8_8(out + t * I + i, v, inv_s, local_max);
+  }
+#pragma unroll
+  for (int off = 1; off < WAVE_SIZE; off <<= 1)
+    local_max = fmaxf(local_max, __shfl_xor(local_max, off, WAVE_SIZE));
+  const int wid = threadIdx.x / WAVE_SIZE;
+  if ((threadIdx.x & (WAVE_SIZE - 1)) == 0) red[wid] = local_max;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = red[0];
+    for (int w2 = 1; w2 < (int)(blockDim.x / WAVE_SIZE); ++w2) m = fmaxf(m, red[w2]);
+    if (m > 0.f)
+      atomicMax(reinterpret_cast<unsigned int*>(amax_next), __float_as_uint(m));
+  }
+}
+
+torch::Tensor swiglu_fp8(torch::Tensor gateup, torch::Tensor scale, torch::Tensor amax_next) {
+  TORCH_CHECK(gateup.is_cuda() && gateup.dtype() == torch::kBFloat16 && gateup.is_contiguous());
+  const int64_t T = gateup.numel() / gateup.size(-1);
+  const int I = (int)gateup.size(-1) / 2;
+  TORCH_CHECK(I % 8 == 0, "swiglu_fp8 needs I % 8 == 0");
+  auto sizes = gateup.sizes().vec();
+  sizes.back() = I;
+  auto y = torch::empty(sizes, gateup.options().dtype(torch::kFloat8_e4m3fn));
+  hipLaunchKernelGGL(swiglu_fp8_kernel8, dim3(grid_for(T * I / 8, 256)), dim3(256), 0,
+                     cur_stream(), (const uint16_t*)gateup.data_ptr(), (uint8_t*)y.data_ptr(),
+                     scale.data_ptr<float>(), amax_next.data_ptr<float>(), T, I);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
 void fp8_scale_update(torch::Tensor scale, torch::Tensor amax_next) {
   const int n = (int)scale.numel();
   TORCH_CHECK(amax_next.numel() == n && scale.is_contiguous() && amax_next.is_contiguous());
