@@ -102,3 +102,68 @@ def test_fp8_linear_delayed_scaling_converges():
     ref = ops.fp8_linear(x, w8, sw, w_bf16=w)
     err = (y2.float() - ref.float()).abs().max().item()
     assert err <= ref.float().abs().max().item() * 0.05 + 0.5, err
+
+
+@requires_gpu
+def test_fused_fp8_producers_match_unfused():
+    """add_rmsnorm_fp8_ / swiglu_fp8 == (bf16 producer -> fp8_quant_delayed)
+    bitwise: same scale, same e4m3 values, same accumulated amax."""
+    torch.manual_seed(9)
+    T, H, I = 64, 512, 1024
+    h1 = (torch.randn(T, H, device="cuda") * 0.5).to(torch.bfloat16)
+    h2 = h1.clone()
+    delta = (torch.randn(T, H, device="cuda") * 0.1).to(torch.bfloat16)
+    w = torch.rand(H, device="cuda").to(torch.bfloat16) + 0.5
+    scale = torch.full((1,), 0.37, device="cuda")
+    amax_a = torch.zeros(1, device="cuda")
+    amax_b = torch.zeros(1, device="cuda")
+
+    # unfused: bf16 add_rmsnorm then standalone delayed quant
+    y_bf16 = ops.add_rmsnorm_(h1, delta.clone(), w, 1e-6)
+    y8_ref = torch.empty(T, H, device="cuda", dtype=torch.float8_e4m3fn)
+    from rllm_amd.ops import require_ext
+
+    require_ext().fp8_quant_delayed(y_bf16.contiguous(), y8_ref, scale, amax_a)
+
+    # fused
+    y8 = ops.add_rmsnorm_fp8_(h2, delta.clone(), w, 1e-6, scale, amax_b)
+    assert torch.equal(h1, h2)  # residual update identical
+    # fused quantizes the fp32 pre-bf16-rounding value; allow 1 e4m3 step
+    d = (y8.float() - y8_ref.float()).abs()
+    assert (d <= (y8_ref.float().abs() * 0.13 + scale.item() * 0.02)).all(), d.max()
+    # amax agrees to bf16 rounding
+    assert abs(amax_a.item() - amax_b.item()) <= abs(amax_a.item()) * 0.01 + 1e-3
+
+    gu = (torch.randn(T, 2 * I, device="cuda") * 0.8).to(torch.bfloat16)
+    out_bf16 = ops.swiglu(gu)
+    amax_a.zero_(), amax_b.zero_()
+    y8_ref2 = torch.empty(T, I, device="cuda", dtype=torch.float8_e4m3fn)
+    require_ext().fp8_quant_delayed(out_bf16.contiguous(), y8_ref2, scale, amax_a)
+    y8_2 = ops.swiglu_fp8(gu, scale, amax_b)
+    d2 = (y8_2.float() - y8_ref2.float()).abs()
+    assert (d2 <= (y8_ref2.float().abs() * 0.13 + scale.item() * 0.02)).all(), d2.max()
+    assert abs(amax_a.item() - amax_b.item()) <= abs(amax_a.item()) * 0.01 + 1e-3
+
+
+@requires_gpu
+def test_engine_fp8_fused_decode_logprob_sanity():
+    """fp8 fused-producer decode stays close to bf16 decode logprobs."""
+    from rllm_amd.engine.inference.llm_engine import LLMEngine, SamplingParams
+    from rllm_amd.models.qwen import QwenModel
+
+    model = QwenModel(CFG, device="cuda").init_random(seed=21)
+    eng_bf16 = LLMEngine(model, kv_budget_bytes=64 << 20, eos_token_id=None, seed=5)
+    prompts = [list(range(5 + i, 37 + i)) for i in range(4)]
+    sp = SamplingParams(temperature=0.0, max_tokens=12)  # greedy for comparability
+    outs_a = eng_bf16.generate(prompts, sp)
+
+    model2 = QwenModel(CFG, device="cuda").init_random(seed=21)
+    eng_fp8 = LLMEngine(model2, kv_budget_bytes=64 << 20, eos_token_id=None,
+                        seed=5, fp8_decode=True)
+    # fused path requires all three producer sites armed
+    assert all(n in model2.layers[0]._fp8 for n in ("qkv_proj", "gate_up_proj", "down_proj"))
+    outs_b = eng_fp8.generate(prompts, sp)
+    agree = sum(int(a == b) for oa, ob in zip(outs_a, outs_b)
+                for a, b in zip(oa.token_ids, ob.token_ids))
+    total = sum(len(o.token_ids) for o in outs_a)
+    assert agree / total > 0.7, f"fp8 fused decode diverged: {agree}/{total} greedy agreement"
