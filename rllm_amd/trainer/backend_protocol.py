@@ -60,6 +60,12 @@ class BackendProtocol(ABC):
         """Called after every optimizer step (weight-sync hook, :196)."""
         return None
 
+    # ---- context curriculum ----
+    def set_max_response_tokens(self, n: int) -> None:
+        """Raise the rollout response cap mid-run (context-length
+        curriculum, DeepScaleR 8K→16K→24K). Default no-op."""
+        return None
+
     # ---- checkpointing ----
     def save_checkpoint(self, path: str, step: int) -> None:
         return None

@@ -85,6 +85,11 @@ class CPUBackend(BackendProtocol):
 
         return shard_rows_balanced(batch, world_size)[rank]
 
+    def set_max_response_tokens(self, n: int) -> None:
+        self.rollout_max_tokens = int(n)
+        if self.flow_engine is not None:
+            self.flow_engine.default_sampling_params["max_tokens"] = int(n)
+
     def update_policy(self, rows: list[PackedRow]) -> dict:
         """Fp32 torch GRPO update (token-mean aggregation)."""
         self.optimizer.zero_grad()

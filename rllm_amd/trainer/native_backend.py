@@ -205,6 +205,11 @@ class NativeBackend(BackendProtocol):
 
         return self.policy.update_policy(rows, old_logprob_fn=old_lp_fn)
 
+    def set_max_response_tokens(self, n: int) -> None:
+        # the engine/flow-engine hold references to this dict — mutating it
+        # raises the cap for every subsequent rollout (context curriculum)
+        self.rollout_sampling_params["max_tokens"] = int(n)
+
     def on_policy_updated(self, weight_version: int) -> None:
         # colocated: engine shares the actor's tensors — only the version
         # stamp moves (cf. reference CheckpointEngineManager.update_weights)

@@ -63,6 +63,11 @@ class TrainerConfig:
     # global-batch + balance_batch semantics, verl/utils.py:310). Off =
     # per-rank shards (each rank trains its own task shard).
     global_batch_mode: bool = False
+    # context-length curriculum (DeepScaleR 8K→16K→24K,
+    # docs/projects/deep-scaler.mdx:20): [[step, max_response_tokens], ...]
+    # — at each listed global step the backend's rollout max_tokens is
+    # raised to the given value
+    context_curriculum: list | None = None
     seed: int = 0
     # rocprofv3-visible step gating (reference trainer.profile_steps):
     # wrap the listed global steps in hipProfilerStart/Stop + roctx ranges
@@ -146,6 +151,10 @@ class UnifiedTrainer:
                     if cfg.max_steps is not None and self.state.global_step >= cfg.max_steps:
                         return
                     t0 = time.monotonic()
+                    if cfg.context_curriculum:
+                        for at_step, max_toks in cfg.context_curriculum:
+                            if self.state.global_step == at_step:
+                                self.backend.set_max_response_tokens(int(max_toks))
                     self.backend.on_batch_start(self.state.global_step)
                     from rllm_amd.utils.profiling import step_profile_region
 

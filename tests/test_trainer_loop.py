@@ -141,3 +141,25 @@ def test_cpu_backend_cumulative_multiturn(tmp_path):
     trainer = UnifiedTrainer(backend, tasks, config=tcfg)
     trainer.fit()
     assert trainer.state.global_step == 1
+
+
+def test_context_curriculum_raises_rollout_cap(tmp_path):
+    """DeepScaleR-style context schedule: the trainer raises the backend's
+    rollout response cap at the configured global steps."""
+    backend = CPUBackend(solve_flow, parity_eval, rollout_max_tokens=4, seed=0)
+    cfg = TrainerConfig(total_epochs=4, train_batch_size=2, rollout_n=2, max_steps=3,
+                        logger_backends=[],
+                        context_curriculum=[[0, 6], [2, 10]])
+    trainer = UnifiedTrainer(backend, make_dataset(2), config=cfg)
+    seen = []
+    orig = backend.set_max_response_tokens
+
+    def spy(n):
+        seen.append((trainer.state.global_step, n))
+        orig(n)
+
+    backend.set_max_response_tokens = spy
+    trainer.fit()
+    assert seen == [(0, 6), (2, 10)]
+    assert backend.rollout_max_tokens == 10
+    assert backend.flow_engine.default_sampling_params["max_tokens"] == 10
