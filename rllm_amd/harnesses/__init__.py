@@ -1,0 +1,24 @@
+from rllm_amd.harnesses import coding_agents as _coding_agents  # registers harnesses
+from rllm_amd.harnesses.cli_harness import (
+    HARNESS_REGISTRY,
+    BaseCliHarness,
+    BashHarness,
+    CurlChatHarness,
+    OracleHarness,
+    get_harness,
+)
+from rllm_amd.harnesses.coding_agents import (
+    AiderHarness,
+    ClaudeCodeHarness,
+    CodexHarness,
+    MiniSweAgentHarness,
+    OpenCodeHarness,
+    QwenCodeHarness,
+)
+
+__all__ = [
+    "HARNESS_REGISTRY", "BaseCliHarness", "get_harness",
+    "BashHarness", "OracleHarness", "CurlChatHarness",
+    "MiniSweAgentHarness", "AiderHarness", "ClaudeCodeHarness",
+    "CodexHarness", "OpenCodeHarness", "QwenCodeHarness",
+]

@@ -48,6 +48,41 @@ class BaseCliHarness:
         """The shell command that runs the agent on the task."""
         raise NotImplementedError
 
+    # -- shared helpers --------------------------------------------------------
+    #: where the agent's stdout is tee'd inside the sandbox (debuggability)
+    stdout_log_path: str | None = None
+
+    @staticmethod
+    def split_model(model: str) -> tuple[str, str, str]:
+        """(provider, bare_name, provider/bare) — litellm-style CLIs insist
+        on a provider prefix; infer one for bare gateway model names."""
+        if "/" in model:
+            provider, bare = model.split("/", 1)
+            return provider, bare, model
+        low = model.lower()
+        if low.startswith("claude"):
+            return "anthropic", model, f"anthropic/{model}"
+        return "openai", model, f"openai/{model}"
+
+    @staticmethod
+    def anthropic_base(url: str) -> str:
+        """The Anthropic SDK appends /v1/messages itself — hand it the
+        gateway session root without the trailing /v1."""
+        u = url.rstrip("/")
+        return u[: -len("/v1")] if u.endswith("/v1") else u
+
+    @staticmethod
+    def workdir(task: Task) -> str | None:
+        md = task.metadata or {}
+        return md.get("workdir") or md.get("repo_dir")
+
+    def cd_prefix(self, task: Task) -> str:
+        wd = self.workdir(task)
+        return f"cd {shlex.quote(wd)} && " if wd else ""
+
+    def tee_suffix(self) -> str:
+        return f" 2>&1 | tee {shlex.quote(self.stdout_log_path)}" if self.stdout_log_path else ""
+
     # -- the flow ------------------------------------------------------------
     async def arun(self, task: Task, config: AgentConfig, *, env: Sandbox) -> None:
         await self.install(env)
