@@ -380,7 +380,10 @@ class QwenModel(nn.Module):
         self._stamp_lora()
         hidden = self.embed_tokens[input_ids]
         flash_tiles = None
-        if self.use_flash_training_attention:
+        # the MFMA training-attention kernel is specialized for D=128 (the
+        # Qwen2.5/R1-Distill head size); other head dims (e.g. 0.5B's 64,
+        # tiny test configs) take the chunked-matmul path
+        if self.use_flash_training_attention and self.cfg.head_dim == 128:
             seqlens = [cu_seqlens[i + 1] - cu_seqlens[i] for i in range(len(cu_seqlens) - 1)]
             flash_tiles = make_prefill_tiles(seqlens, hidden.device)
         for layer in self.layers:

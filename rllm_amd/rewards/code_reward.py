@@ -64,6 +64,31 @@ print("RESULTS:" + json.dumps(results))
 """
 
 
+import functools
+import os
+import shutil
+
+
+@functools.lru_cache(maxsize=1)
+def _unshare_prefix() -> list[str]:
+    """Firejail-class isolation for untrusted model code (reference
+    code_utils/firejail_exec.py): detach network + PID namespace via
+    util-linux `unshare` when the host allows it (root / user namespaces).
+    Falls back to the plain rlimited subprocess otherwise."""
+    exe = shutil.which("unshare")
+    if not exe:
+        return []
+    probe = subprocess.run([exe, "-n", "-p", "--fork", "true"],
+                           capture_output=True, timeout=10)
+    if probe.returncode == 0:
+        return [exe, "-n", "-p", "--fork"]
+    probe = subprocess.run([exe, "-r", "-n", "-p", "--fork", "true"],
+                           capture_output=True, timeout=10)
+    if probe.returncode == 0:  # rootless via user namespace
+        return [exe, "-r", "-n", "-p", "--fork"]
+    return []
+
+
 def run_tests(code: str, tests: list[dict], timeout: float = 10.0) -> CodeRewardOutput:
     """tests: [{"stdin": ..., "stdout": ...}] or [{"assert": "assert f(2)==4"}]."""
     if not tests:
@@ -71,10 +96,12 @@ def run_tests(code: str, tests: list[dict], timeout: float = 10.0) -> CodeReward
     with tempfile.TemporaryDirectory() as td:
         spec = Path(td) / "spec.json"
         spec.write_text(json.dumps({"code": code, "tests": tests}))
+        env = {"PATH": os.environ.get("PATH", "/usr/bin:/bin"), "HOME": td,
+               "TMPDIR": td, "PYTHONDONTWRITEBYTECODE": "1"}
         try:
             proc = subprocess.run(
-                [sys.executable, "-I", "-c", _RUNNER, str(spec)],
-                capture_output=True, text=True, timeout=timeout, cwd=td)
+                [*_unshare_prefix(), sys.executable, "-I", "-c", _RUNNER, str(spec)],
+                capture_output=True, text=True, timeout=timeout, cwd=td, env=env)
         except subprocess.TimeoutExpired:
             return CodeRewardOutput(0.0, False, 0, len(tests), error="timeout")
     m = re.search(r"RESULTS:(\[.*\])", proc.stdout)

@@ -134,3 +134,27 @@ def llm_judge_reward_fn(task, episode) -> float:
         _default_judge = LLMJudgeReward(
             base_url=url, model=os.environ.get("RLLM_JUDGE_MODEL", "judge"))
     return _default_judge(task, episode)
+
+
+def make_equality_judge(base_url: str, model: str, timeout: float = 60.0):
+    """Binary semantic-equality judge for llm_equality_reward (reference
+    llm_equality.py judge stage): returns a callable
+    (candidate, reference) -> bool | None (None = judge unavailable)."""
+    chat = http_chat_fn(base_url, model, timeout=timeout)
+
+    def judge(candidate: str, reference: str):
+        import json as _json
+
+        prompt = (
+            "You are an impartial judge. Is the candidate answer semantically "
+            "equivalent to the reference answer? Numbers must match in value. "
+            'Respond with JSON only: {"equivalent": true} or {"equivalent": false}.\n\n'
+            f"Reference answer: {reference}\n\nCandidate answer: {candidate}")
+        try:
+            out = chat([{"role": "user", "content": prompt}])
+            m = out[out.index("{"): out.rindex("}") + 1]
+            return bool(_json.loads(m).get("equivalent"))
+        except Exception:  # noqa: BLE001 — unreachable judge -> fallback path
+            return None
+
+    return judge

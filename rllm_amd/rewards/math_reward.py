@@ -105,17 +105,56 @@ def _time_limit(seconds: float):
         signal.signal(signal.SIGALRM, old)
 
 
-def _sympy_equal(a: str, b: str, timeout: float = 3.0) -> bool:
-    try:
-        with _time_limit(timeout):
-            import sympy
-            from sympy.parsing.sympy_parser import parse_expr, standard_transformations, implicit_multiplication_application
+def _sympy_equal_inproc(a: str, b: str) -> bool:
+    import sympy
+    from sympy.parsing.sympy_parser import parse_expr, standard_transformations, implicit_multiplication_application
 
-            tf = standard_transformations + (implicit_multiplication_application,)
-            ea = parse_expr(a.replace("^", "**"), transformations=tf)
-            eb = parse_expr(b.replace("^", "**"), transformations=tf)
-            return bool(sympy.simplify(ea - eb) == 0)
-    except Exception:  # noqa: BLE001 — parse failures / timeouts mean "not equal"
+    tf = standard_transformations + (implicit_multiplication_application,)
+    ea = parse_expr(a.replace("^", "**"), transformations=tf)
+    eb = parse_expr(b.replace("^", "**"), transformations=tf)
+    return bool(sympy.simplify(ea - eb) == 0)
+
+
+# Process pool for sympy equivalence: SIGALRM cannot interrupt C-level
+# sympy loops, so a stuck simplify() would hang the training step. A child
+# process gets a HARD timeout (killed + pool rebuilt). Lazy singleton —
+# the reference's sympy-timeout goal done the robust way.
+_SYMPY_POOL = None
+
+
+def _get_sympy_pool():
+    global _SYMPY_POOL
+    if _SYMPY_POOL is None:
+        import concurrent.futures as cf
+        import multiprocessing as mp
+
+        _SYMPY_POOL = cf.ProcessPoolExecutor(
+            max_workers=1, mp_context=mp.get_context("spawn"))
+    return _SYMPY_POOL
+
+
+def _sympy_equal(a: str, b: str, timeout: float = 3.0) -> bool:
+    global _SYMPY_POOL
+    try:
+        pool = _get_sympy_pool()
+    except Exception:  # noqa: BLE001 — no subprocess allowed: SIGALRM fallback
+        try:
+            with _time_limit(timeout):
+                return _sympy_equal_inproc(a, b)
+        except Exception:  # noqa: BLE001
+            return False
+    try:
+        fut = pool.submit(_sympy_equal_inproc, a, b)
+        return bool(fut.result(timeout=timeout))
+    except Exception:  # noqa: BLE001 — timeout / parse failure / broken pool
+        # a timed-out child keeps burning CPU — kill and rebuild next call
+        pool.shutdown(wait=False, cancel_futures=True)
+        for p in getattr(pool, "_processes", {}).values():
+            try:
+                p.terminate()
+            except Exception:  # noqa: BLE001
+                pass
+        _SYMPY_POOL = None
         return False
 
 

@@ -14,10 +14,14 @@ prompt. Reports rollout/update wall time, tokens/s, and HBM headroom.
 from __future__ import annotations
 
 import argparse
+import sys
 import time
+from pathlib import Path
 
 import numpy as np
 import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
 
 def main():
