@@ -50,15 +50,41 @@ class FixedEvaluatorHooks:
         return None
 
 
+class SandboxVerifierEvaluator:
+    """Per-task verifier that runs INSIDE the task's sandbox (the SWE-bench
+    builder's tests/test.sh contract, data/swe_builders.py): exit 0 =
+    solved. Created by SandboxTaskHooks when the task declares a
+    [verifier] command (reference hooks.py:245 verifier resolution)."""
+
+    TASK_MOUNT = "/rllm_task"
+
+    def __init__(self, env, command: str, timeout: float = 1800.0):
+        self.env = env
+        self.command = command
+        self.timeout = timeout
+
+    async def aevaluate(self, task, episode):
+        from rllm_amd.eval.types import EvalOutput
+
+        res = await self.env.exec(self.command, timeout=self.timeout, cwd=self.TASK_MOUNT)
+        ok = res.ok
+        return EvalOutput(reward=1.0 if ok else 0.0, is_correct=ok,
+                          signals={"verifier_exit": float(res.exit_code)})
+
+
 class SandboxTaskHooks:
     """Provisions a sandbox per task and resolves the task's verifier
-    (reference hooks.py:201-293). The sandbox manager supplies acquire/
-    release; evaluator resolution falls back to ``fixed_evaluator``."""
+    (reference hooks.py:201-293). A task whose metadata carries
+    ``[verifier] command`` (the task.toml shape load_tasks produces) gets a
+    SandboxVerifierEvaluator bound to its sandbox, with the task dir's
+    tests/ uploaded; otherwise ``fixed_evaluator`` applies."""
 
-    def __init__(self, sandbox_manager=None, fixed_evaluator=None, warm_queue=None):
+    def __init__(self, sandbox_manager=None, fixed_evaluator=None, warm_queue=None,
+                 verifier_timeout: float = 1800.0):
         self.sandbox_manager = sandbox_manager
         self.fixed_evaluator = fixed_evaluator
         self.warm_queue = warm_queue
+        self.verifier_timeout = verifier_timeout
 
     async def setup(self, task: Any, uid: str) -> TaskContext:
         env = None
@@ -68,7 +94,32 @@ class SandboxTaskHooks:
             else:
                 env = await self.sandbox_manager.acquire(task)
         evaluator = self.fixed_evaluator
+        md = getattr(task, "metadata", None) or {}
+        ver = md.get("verifier") if isinstance(md, dict) else None
+        cmd = ver.get("command") if isinstance(ver, dict) else None
+        if env is not None and cmd:
+            await self._upload_task_files(task, env)
+            evaluator = SandboxVerifierEvaluator(env, cmd, timeout=self.verifier_timeout)
         return TaskContext(evaluator=evaluator, env=env)
+
+    async def _upload_task_files(self, task: Any, env) -> None:
+        """Copy the task directory's verifier fixtures (tests/ etc.) into
+        the sandbox under /rllm_task so the verifier command can run with
+        that cwd."""
+        from pathlib import Path
+
+        base = getattr(task, "dataset_dir", None)
+        sub = getattr(task, "sub_dir", None)
+        if not base or sub is None:
+            return
+        task_dir = Path(base) / sub
+        if not task_dir.is_dir():
+            return
+        mount = SandboxVerifierEvaluator.TASK_MOUNT
+        for f in sorted(task_dir.rglob("*")):
+            if f.is_file() and "solution" not in f.parts:  # never ship the oracle
+                rel = f.relative_to(task_dir)
+                await env.upload(str(f), f"{mount}/{rel}")
 
     async def teardown(self, task: Any, uid: str, ctx: TaskContext) -> None:
         if ctx.env is not None and self.sandbox_manager is not None:

@@ -57,6 +57,13 @@ class LocalSandbox:
         full_env = dict(os.environ)
         if env:
             full_env.update(env)
+        if cwd:
+            # mirror upload/write_file semantics: absolute container-style
+            # paths map under the scratch workdir when they don't exist
+            p = Path(cwd)
+            if p.is_absolute() and not p.exists():
+                p = self.workdir / cwd.lstrip("/")
+            cwd = str(p)
         return await _run(command, timeout, cwd=cwd or str(self.workdir), env=full_env, shell=True)
 
     async def upload(self, local_path: str, remote_path: str) -> None:
