@@ -21,6 +21,10 @@ class LazyGroup(click.Group):
         "dataset": "rllm_amd.cli.dataset_cmd:dataset_cmd",
         "view": "rllm_amd.cli.view:view",
         "serve": "rllm_amd.cli.serve:serve",
+        "sft": "rllm_amd.cli.sft_cmd:sft",
+        "model": "rllm_amd.cli.model_cmd:model_cmd",
+        "init": "rllm_amd.cli.init_cmd:init_cmd",
+        "snapshot": "rllm_amd.cli.snapshot_cmd:snapshot_cmd",
     }
 
     def list_commands(self, ctx):
