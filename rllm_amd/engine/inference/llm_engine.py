@@ -520,9 +520,16 @@ class LLMEngine:
     # ------------------------------------------------------------------
     _GRAPH_SIZES = [1, 2, 4, 8, 16, 32, 48, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
 
+    #: max decode steps per graph block (in-graph state advance between
+    #: replays: no host work, no H2D staging, no D2H sync inside a block)
+    BLOCK_MAX = 16
+    #: block length when stop tokens are live — bounds over-decode past EOS
+    EOS_BLOCK = 4
+
     def _init_graph_buffers(self):
         dev = self.device
         Bmax = min(self.max_num_seqs, 1024)
+        K = self.BLOCK_MAX
         self._gb = {
             "tok": torch.zeros(Bmax, device=dev, dtype=torch.long),
             "pos": torch.zeros(Bmax, device=dev, dtype=torch.int32),
@@ -530,21 +537,40 @@ class LLMEngine:
             "bt": torch.zeros(Bmax, self._bt_width, device=dev, dtype=torch.int32),
             "len": torch.ones(Bmax, device=dev, dtype=torch.int32),
             "step": torch.zeros(1, device=dev, dtype=torch.int32),
+            # block machinery: per-step slot schedule, in-block index, and
+            # token/logprob history read back ONCE per block
+            "slot_sched": torch.zeros(K, Bmax, device=dev, dtype=torch.int32),
+            "blk_i": torch.zeros(1, device=dev, dtype=torch.long),
+            "tok_hist": torch.zeros(K, Bmax, device=dev, dtype=torch.long),
+            "lp_hist": torch.zeros(K, Bmax, device=dev, dtype=torch.float32),
         }
         self._hb = {
             "tok": torch.zeros(Bmax, dtype=torch.long, pin_memory=True),
             "pos": torch.zeros(Bmax, dtype=torch.int32, pin_memory=True),
-            "slot": torch.zeros(Bmax, dtype=torch.int32, pin_memory=True),
+            "slot_sched": torch.zeros(K, Bmax, dtype=torch.int32, pin_memory=True),
             "bt": torch.zeros(Bmax, self._bt_width, dtype=torch.int32, pin_memory=True),
             "len": torch.ones(Bmax, dtype=torch.int32, pin_memory=True),
         }
 
     def _decode_compute(self, Bp: int, temp: float):
+        """One decode step INCLUDING the state advance for the next step —
+        a block of R steps is R back-to-back replays with zero host work."""
         g = self._gb
+        g["step"].add_(1)
         hidden = self.model.forward_decode(g["tok"][:Bp], g["pos"][:Bp], self.kv,
                                            g["slot"][:Bp], g["bt"][:Bp], g["len"][:Bp])
         logits = self.model.logits(hidden).contiguous()
         tokens, lps = ops.sample_logprob(logits, temp, self.seed, 0, step_tensor=g["step"])
+        i = g["blk_i"]
+        g["tok_hist"][:, :Bp].index_copy_(0, i, tokens.unsqueeze(0))
+        g["lp_hist"][:, :Bp].index_copy_(0, i, lps.unsqueeze(0))
+        # advance: next step's inputs (sampled token, pos+1, next slot row)
+        g["tok"][:Bp].copy_(tokens)
+        g["pos"][:Bp].add_(1)
+        g["len"][:Bp].add_(1)
+        g["blk_i"].add_(1)
+        nxt = torch.remainder(g["blk_i"], self.BLOCK_MAX)
+        g["slot"][:Bp].copy_(g["slot_sched"].index_select(0, nxt)[0, :Bp])
         return tokens, lps
 
     def _get_graph(self, Bp: int, temp: float):
@@ -575,27 +601,44 @@ class LLMEngine:
         self._graphs[key] = (graph, tokens, lps)
         return self._graphs[key]
 
+    def _block_len(self, batch: list[Sequence]) -> int:
+        """Steps to run without a host check: bounded by max_tokens /
+        context headroom per seq, and by EOS_BLOCK when stop tokens are
+        live (over-decode past a stop is computed but never appended)."""
+        R = self.BLOCK_MAX
+        if self.eos_token_id is not None or any(s.params.stop_token_ids for s in batch):
+            R = self.EOS_BLOCK
+        for s in batch:
+            R = min(R,
+                    s.params.max_tokens - len(s.output_ids),
+                    self.max_model_len - s.total_len)
+        return max(1, R)
+
     def _run_decode_graph(self, batch: list[Sequence], temp: float) -> int:
         if self._gb is None:
             self._init_graph_buffers()
         B = len(batch)
         Bp = next(s for s in self._GRAPH_SIZES if s >= B)
         hb, gb = self._hb, self._gb
+        R = self._block_len(batch)
 
         tok_np = hb["tok"].numpy()
         pos_np = hb["pos"].numpy()
-        slot_np = hb["slot"].numpy()
+        sched_np = hb["slot_sched"].numpy()
         len_np = hb["len"].numpy()
         bt_np = hb["bt"].numpy()
         for i, seq in enumerate(batch):
             if seq.state != SeqState.RUNNING:
                 return 0  # preempted mid-loop; retry next step with new batch
             pos = seq.total_len
-            if not self._grow_pages(seq, pos):
+            # reserve pages for the WHOLE block up front (no host work mid-block)
+            if not self._grow_pages(seq, pos + R - 1):
                 return 0  # batch composition changed; retry next step
             tok_np[i] = seq.last_token
             pos_np[i] = pos
-            slot_np[i] = seq.pages[pos // PAGE_SIZE] * PAGE_SIZE + pos % PAGE_SIZE
+            for r in range(R):
+                p = pos + r
+                sched_np[r, i] = seq.pages[p // PAGE_SIZE] * PAGE_SIZE + p % PAGE_SIZE
             len_np[i] = pos + 1
             np_pages = seq.pages
             bt_np[i, : len(np_pages)] = np_pages
@@ -603,26 +646,41 @@ class LLMEngine:
         if Bp > B:
             tok_np[B:Bp] = 0
             pos_np[B:Bp] = 0
-            slot_np[B:Bp] = 0
+            sched_np[:, B:Bp] = 0
             len_np[B:Bp] = 1
             bt_np[B:Bp, 0] = 0
 
-        for k in ("tok", "pos", "slot", "len"):
-            gb[k][:Bp].copy_(hb[k][:Bp], non_blocking=True)
-        gb["bt"][:Bp].copy_(hb["bt"][:Bp], non_blocking=True)
-        gb["step"].add_(1)
+        def stage():
+            for k in ("tok", "pos", "len"):
+                gb[k][:Bp].copy_(hb[k][:Bp], non_blocking=True)
+            gb["bt"][:Bp].copy_(hb["bt"][:Bp], non_blocking=True)
+            gb["slot_sched"][:, :Bp].copy_(hb["slot_sched"][:, :Bp], non_blocking=True)
+            gb["blk_i"].zero_()
+            gb["slot"][:Bp].copy_(gb["slot_sched"][0, :Bp])
 
-        graph, tokens_out, lps_out = self._get_graph(Bp, temp)
-        graph.replay()
+        stage()
+        fresh = (Bp, temp) not in self._graphs
+        graph, _tokens_out, _lps_out = self._get_graph(Bp, temp)
+        if fresh:
+            # capture ran the compute (incl. state advance) on live buffers —
+            # restage so replay 0 starts from this block's real inputs
+            stage()
+        for _ in range(R):
+            graph.replay()
 
-        tokens = tokens_out[:B].tolist()
-        logprobs = lps_out[:B].tolist()
-        self._sample_step += 1
-        for seq, tok, lp in zip(batch, tokens, logprobs):
-            seq.output_ids.append(int(tok))
-            seq.logprobs.append(float(lp))
-            self._maybe_finish(seq)
-        return B
+        tok_hist = gb["tok_hist"][:R, :B].tolist()
+        lp_hist = gb["lp_hist"][:R, :B].tolist()
+        self._sample_step += R
+        n_tokens = 0
+        for i, seq in enumerate(batch):
+            for r in range(R):
+                if seq.state != SeqState.RUNNING:
+                    break  # finished mid-block: later tokens computed but dropped
+                seq.output_ids.append(int(tok_hist[r][i]))
+                seq.logprobs.append(float(lp_hist[r][i]))
+                n_tokens += 1
+                self._maybe_finish(seq)
+        return n_tokens
 
     # ------------------------------------------------------------------
     def _sample_and_append(self, batch: list[Sequence], hidden: torch.Tensor) -> None:

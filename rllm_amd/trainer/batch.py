@@ -153,12 +153,16 @@ def rows_from_episodes(episodes: list[Episode]) -> list[PackedRow]:
     return rows
 
 
-def pack_rows(rows: list[PackedRow], device: str = "cpu", max_seq_len: int | None = None) -> TrainBatch:
+def pack_rows(rows: list[PackedRow], device: str = "cpu", max_seq_len: int | None = None,
+              pad_to_multiple: int | None = None) -> TrainBatch:
     """Pack rows into one varlen batch.
 
     Shift semantics: the logprob of token t is computed at row t-1, so
     loss_mask[r] = response_mask[r+1] and targets[r] = tokens[r+1]
     (within each sequence; a sequence's last row is never a loss row).
+
+    pad_to_multiple: append a dummy fully-masked sequence so the packed
+    length divides the Ulysses SP degree (parallel/ulysses.py shard_slice).
     """
     input_ids: list[int] = []
     positions: list[int] = []
