@@ -196,6 +196,18 @@ def pack_rows(rows: list[PackedRow], device: str = "cpu", max_seq_len: int | Non
                 advantages.append(0.0)
                 rollout_lps.append(0.0)
 
+    if pad_to_multiple and len(input_ids) % pad_to_multiple:
+        pad = pad_to_multiple - len(input_ids) % pad_to_multiple
+        if pad == 1:  # a 1-token sequence is degenerate; pad a full extra round
+            pad += pad_to_multiple
+        input_ids.extend([0] * pad)
+        positions.extend(range(pad))
+        cu.append(cu[-1] + pad)
+        loss_mask.extend([False] * pad)
+        targets.extend([0] * pad)
+        advantages.extend([0.0] * pad)
+        rollout_lps.extend([0.0] * pad)
+
     return TrainBatch(
         input_ids=torch.tensor(input_ids, dtype=torch.long, device=device),
         positions=torch.tensor(positions, dtype=torch.int32, device=device),

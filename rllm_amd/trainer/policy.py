@@ -60,6 +60,10 @@ class PolicyTrainerConfig:
     # run the frozen-ref no-grad forward of micro i+1 on a second HIP
     # stream, overlapped with micro i's backward (single-GPU latency hide)
     overlap_ref_stream: bool = True
+    # Ulysses sequence-parallel update (SURVEY §2.D SP; parallel/ulysses.py):
+    # ALL ranks hold the same rows; each runs its token shard of every
+    # micro with all-to-alls around attention. Long-context updates only.
+    sequence_parallel: bool = False
     use_ref: bool = True
     # LoRA training: the KL reference is the actor's own base weights with
     # adapters disabled (models/lora.py) — no second model copy.
@@ -152,6 +156,8 @@ class PolicyTrainer:
         fused AdamW. Returns metrics.
         """
         cfg = self.cfg
+        if cfg.sequence_parallel and pdist.get_world_size() > 1:
+            return self._update_policy_sp(rows)
         micros = split_rows_token_balanced(rows, cfg.max_tokens_per_micro)
         # DP desync guard: every rank must run the same number of micro
         # fwd/bwd? No — grads accumulate locally; only the all-reduce at the
@@ -310,6 +316,119 @@ class PolicyTrainer:
         }
         metrics.update(offpolicy_metrics)
         return metrics
+
+    # ------------------------------------------------------------------
+    def _update_policy_sp(self, rows: list[PackedRow]) -> dict:
+        """Ulysses SP update (trainer-side plumbing, TODO r1 #10): every
+        rank holds the SAME rows; each packs the same micros (padded to the
+        SP degree) and runs forward/backward on ITS token shard with
+        all-to-alls around attention (models/qwen.py sp_group). Gradients
+        are partial sums over disjoint loss-token shards → the same
+        all-reduce(SUM) as DP completes them.
+
+        old-logprob source must be 'alias' or 'rollout' here (a recompute
+        callback would need its own sharded forward)."""
+        import torch.distributed as tdist
+
+        from rllm_amd.parallel import ulysses
+
+        cfg = self.cfg
+        if cfg.old_logprob_mode == "recompute" and not cfg.bypass_mode:
+            raise ValueError("sequence_parallel update needs old_logprob_mode "
+                             "'alias' or 'rollout' (no recompute callback)")
+        if cfg.loss_agg_mode != "token-mean":
+            raise ValueError("sequence_parallel update supports token-mean only")
+        P = pdist.get_world_size()
+        rank = pdist.get_rank()
+        group = self.model.sp_group or tdist.group.WORLD
+        self.model.sp_group = group
+        if self.ref_model is not None:
+            self.ref_model.sp_group = group
+
+        micros = split_rows_token_balanced(rows, cfg.max_tokens_per_micro)
+        # C3 (reference patch.py:23-66): attention all-to-alls run INSIDE
+        # each micro — ranks MUST agree on the micro count or the group
+        # deadlocks. Rows are replicated so the schedules match; verify
+        # with an all_reduce(MAX) before entering the loop.
+        n_micro_max = pdist.all_reduce_scalar(float(len(micros)), op="max")
+        if n_micro_max != float(len(micros)):
+            raise RuntimeError(
+                f"SP micro-batch schedule diverged: local {len(micros)} vs max {n_micro_max} "
+                "(rows must be identical on every SP rank)")
+
+        # global denominator: rows are the FULL batch on every rank
+        n_total_tokens = float(sum(sum(r.response_mask) for r in rows))
+        if n_total_tokens <= 0:
+            return {"actor/skipped": 1.0}
+
+        self.optim.zero_grad()
+        tot_loss = 0.0
+        tot_clip = 0.0
+        tot_kl_count = 0
+        ent_sum, ent_n = 0.0, 0
+        n_local_tokens = 0
+
+        eps_hi = cfg.eps_clip_high if cfg.eps_clip_high is not None else cfg.eps_clip
+        for micro in micros:
+            batch = pack_rows(micro, device=str(self.device), pad_to_multiple=P)
+            T = batch.input_ids.shape[0]
+            sl = ulysses.shard_slice(T, rank, P)
+            hidden = self.model.forward_train(batch.input_ids[sl], batch.positions[sl],
+                                              batch.cu_seqlens)
+            lm = batch.loss_mask[sl]
+            rows_idx = lm.nonzero(as_tuple=True)[0]
+            if rows_idx.numel() == 0:
+                # this shard holds no loss tokens — backward still must run
+                # (the all-to-alls have gradient paths on every rank)
+                loss = (hidden.float() * 0).sum()
+                loss.backward()
+                continue
+            tgt = batch.targets[sl][rows_idx]
+            adv = batch.advantages[sl][rows_idx]
+            rollout_lp = batch.rollout_logprobs[sl][rows_idx]
+            lp, ent = ops.chunked_logprob(hidden[rows_idx], self.model.lm_weight, tgt,
+                                          chunk=cfg.entropy_chunk,
+                                          temperature=cfg.temperature, want_entropy=True)
+            if ent is not None:
+                ent_sum += float(ent.sum())
+                ent_n += ent.numel()
+            with torch.no_grad():
+                old_lp = rollout_lp if (cfg.bypass_mode or cfg.old_logprob_mode == "rollout") \
+                    else lp.detach()
+                ref_lp = None
+                if cfg.use_ref and self.ref_model is not None:
+                    rh = self.ref_model.forward_train(batch.input_ids[sl], batch.positions[sl],
+                                                      batch.cu_seqlens)
+                    ref_lp, _ = ops.chunked_logprob(rh[rows_idx], self.ref_model.lm_weight,
+                                                    tgt, chunk=cfg.entropy_chunk,
+                                                    temperature=cfg.temperature,
+                                                    want_entropy=False)
+                tis_w = self._tis_weights(old_lp, rollout_lp)
+            loss_tok, clipped = ops.grpo_loss_per_token(
+                lp, old_lp, ref_lp, adv, tis_w,
+                eps_lo=cfg.eps_clip, eps_hi=eps_hi, kl_beta=cfg.kl_beta)
+            loss = loss_tok.sum() / n_total_tokens
+            loss.backward()
+            tot_loss += float(loss.detach())
+            tot_clip += float(clipped.sum())
+            tot_kl_count += int(loss_tok.numel())
+            n_local_tokens += int(rows_idx.numel())
+
+        # shard grads are partial sums over disjoint loss tokens
+        pdist.all_reduce_sum_(self.flat_grad)
+        gnorm_sq = self.optim.step(grad_scale=1.0)
+        self.weight_version += 1
+        loss_global = pdist.all_reduce_scalar(tot_loss, op="sum")
+        return {
+            "actor/loss": loss_global,
+            "actor/clipfrac": tot_clip / max(1, tot_kl_count),
+            "actor/entropy": ent_sum / max(1, ent_n),
+            "actor/grad_norm": float(gnorm_sq.sqrt().item()),
+            "actor/lr": self.optim.lr_at(self.optim.step_count - 1),
+            "actor/n_micro_batches": len(micros),
+            "actor/n_response_tokens": n_local_tokens,
+            "actor/sp_degree": P,
+        }
 
     # ------------------------------------------------------------------
     def state_dict(self) -> dict:

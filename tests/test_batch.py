@@ -100,3 +100,19 @@ def test_max_seq_len_truncation():
     batch = pack_rows(rows, max_seq_len=15)
     assert batch.input_ids.numel() == 15
     assert batch.n_response_tokens == 5  # tokens 10..14 are responses (shifted)
+
+
+def test_pack_rows_pad_to_multiple():
+    from rllm_amd.trainer.batch import PackedRow, pack_rows
+
+    rows = [PackedRow(tokens=[1, 2, 3, 4, 5], response_mask=[0, 0, 1, 1, 1],
+                      advantages=[0.5] * 5, rollout_logprobs=[-1.0] * 5)]
+    for P in (2, 3, 4, 8):
+        b = pack_rows(rows, pad_to_multiple=P)
+        assert b.input_ids.shape[0] % P == 0, P
+        # pad tokens are never loss rows and never counted
+        assert b.n_response_tokens == 3
+        assert not b.loss_mask[b.cu_seqlens[-2]:].any()
+    # already divisible: no pad sequence added
+    b = pack_rows(rows, pad_to_multiple=5)
+    assert b.input_ids.shape[0] == 5 and b.n_rows == 1
