@@ -36,6 +36,12 @@ class BackendProtocol(ABC):
     def transform_to_backend_batch(self, groups: list[TrajectoryGroup]) -> Any:
         ...
 
+    def postprocess_episodes(self, episodes: list[Episode]) -> dict:
+        """Optional stage between generation and transform — e.g. on-policy
+        distillation writes teacher-KL advantages here (reference
+        agent_workflow_trainer.py:704-766). Returns metrics."""
+        return {}
+
     def process_backend_batch(self, batch: Any) -> Any:
         """Old-logprob / ref-logprob computation etc.; default no-op."""
         return batch

@@ -31,7 +31,7 @@ class CPUBackend(BackendProtocol):
                  lr: float = 1e-4, eps_clip: float = 0.2, kl_beta: float = 0.0,
                  n_parallel_tasks: int = 16, rollout_max_tokens: int = 16,
                  temperature: float = 1.0, seed: int = 0,
-                 gateway_config: GatewayConfig | None = None):
+                 gateway_config: GatewayConfig | None = None, distill=None):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
         self.model = model or TinyTorchLM(seed=seed)
@@ -54,6 +54,7 @@ class CPUBackend(BackendProtocol):
         self.flow_engine: AgentFlowEngine | None = None
         self.weight_version_ref = {"v": 0}
         self.gateway_config = gateway_config
+        self.distill = distill
 
     # ------------------------------------------------------------------
     def init_rollout_engine(self):
@@ -84,6 +85,16 @@ class CPUBackend(BackendProtocol):
         from rllm_amd.trainer.batch import shard_rows_balanced
 
         return shard_rows_balanced(batch, world_size)[rank]
+
+    def postprocess_episodes(self, episodes):
+        if self.distill is None:
+            return {}
+        from rllm_amd.trainer.distill import TeacherClient, distill_episodes
+
+        if isinstance(self.distill, TeacherClient):
+            return distill_episodes(episodes, self.distill)
+        d = dict(self.distill)
+        return distill_episodes(episodes, d.pop("teacher"), **d)
 
     def set_max_response_tokens(self, n: int) -> None:
         self.rollout_max_tokens = int(n)

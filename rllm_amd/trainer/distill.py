@@ -72,3 +72,72 @@ def align_cross_tokenizer(student_tokens: list[str], teacher_tokens: list[str],
                 break
         out[si] = acc
     return out.tolist()
+
+
+# ---------------------------------------------------------------------------
+# Teacher logprob fetch + trainer wiring (reference
+# agent_workflow_trainer.py:704-766: teacher logprobs from an
+# OpenAI-compatible endpoint inside the training loop)
+# ---------------------------------------------------------------------------
+
+
+class TeacherClient:
+    """Fetch per-token logprobs for realized sequences from an
+    OpenAI-compatible /v1/completions endpoint (echo mode: prompt=tokens,
+    max_tokens=0, echo+logprobs return the prompt's own logprobs).
+
+    `transport` overrides HTTP for tests: callable(list[int]) ->
+    list[float | None] (one entry per token; position 0 is None — no
+    context)."""
+
+    def __init__(self, base_url: str = "", model: str = "", transport=None,
+                 timeout: float = 120.0):
+        self.base_url = base_url.rstrip("/")
+        self.model = model
+        self.transport = transport
+        self.timeout = timeout
+
+    def prompt_logprobs(self, token_ids: list[int]) -> list:
+        if self.transport is not None:
+            return self.transport(token_ids)
+        import httpx
+
+        r = httpx.post(f"{self.base_url}/completions",
+                       json={"model": self.model, "prompt": token_ids,
+                             "max_tokens": 0, "echo": True, "logprobs": 1},
+                       timeout=self.timeout)
+        r.raise_for_status()
+        body = r.json()
+        return body["choices"][0]["logprobs"]["token_logprobs"]
+
+    def response_logprobs(self, prompt_ids: list[int], response_ids: list[int]) -> list[float]:
+        """Teacher logprobs for just the response tokens."""
+        full = list(prompt_ids) + list(response_ids)
+        lps = self.prompt_logprobs(full)
+        out = lps[len(prompt_ids):]
+        return [0.0 if v is None else float(v) for v in out]
+
+
+def distill_episodes(episodes: list[Episode], teacher: TeacherClient, *,
+                     coef: float = 1.0, clip: float = 5.0, gamma: float = 0.0) -> dict:
+    """The on-policy distillation stage: per step, fetch teacher logprobs
+    for the realized response and write reverse-KL advantages; episodes
+    then flow through the precomputed-advantage estimator
+    (AlgorithmConfig.use_precomputed_advantage=True)."""
+    n_steps = 0
+    kl_sum = 0.0
+    n_tok = 0
+    for ep in episodes:
+        per_step = []
+        for traj in ep.trajectories:
+            for step in traj.steps:
+                t_lp = teacher.response_logprobs(step.prompt_ids, step.response_ids)
+                per_step.append(t_lp)
+                n_steps += 1
+                kl_sum += float(np.sum(np.asarray(step.logprobs) - np.asarray(t_lp)))
+                n_tok += len(t_lp)
+        write_distill_advantages(ep, per_step, coef=coef, clip=clip, gamma=gamma)
+    return {
+        "distill/steps": float(n_steps),
+        "distill/kl_student_teacher": kl_sum / max(1, n_tok),
+    }

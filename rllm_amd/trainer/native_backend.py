@@ -62,6 +62,7 @@ class NativeBackend(BackendProtocol):
         lora=None,  # models.lora.LoRAConfig | True for defaults | None = full finetune
         gateway_config=None,  # GatewayConfig; e.g. cumulative_token_mode=True for multi-turn
         fp8_rollout: bool = False,  # e4m3 decode GEMMs (wins at >=14B; profiles/)
+        distill=None,  # trainer.distill.TeacherClient | dict(teacher=..., coef=, clip=, gamma=)
     ):
         self.agent_flow = agent_flow
         self.evaluator = evaluator
@@ -104,6 +105,7 @@ class NativeBackend(BackendProtocol):
         self.lora_config = lora
         self.gateway_config = gateway_config
         self.fp8_rollout = fp8_rollout
+        self.distill = distill
 
         self.model: QwenModel | None = None
         self.ref_model: QwenModel | None = None
@@ -204,6 +206,20 @@ class NativeBackend(BackendProtocol):
                 return lp
 
         return self.policy.update_policy(rows, old_logprob_fn=old_lp_fn)
+
+    def postprocess_episodes(self, episodes: list[Episode]) -> dict:
+        """On-policy distillation (reference agent_workflow_trainer.py:704-766):
+        fetch teacher logprobs and write reverse-KL advantages; the
+        precomputed-advantage estimator passes them through."""
+        if self.distill is None:
+            return {}
+        from rllm_amd.trainer.distill import TeacherClient, distill_episodes
+
+        if isinstance(self.distill, TeacherClient):
+            return distill_episodes(episodes, self.distill)
+        d = dict(self.distill)
+        teacher = d.pop("teacher")
+        return distill_episodes(episodes, teacher, **d)
 
     def set_max_response_tokens(self, n: int) -> None:
         # the engine/flow-engine hold references to this dict — mutating it

@@ -188,6 +188,9 @@ class UnifiedTrainer:
         metrics["time/gen_s"] = time.monotonic() - t
         self.state.total_episodes += len(episodes)
         metrics.update(_termination_metrics(episodes))
+        # optional S1.5 — backend episode postprocessing (distillation
+        # teacher-logprob fetch, reward shaping, ...)
+        metrics.update(self.backend.postprocess_episodes(episodes) or {})
 
         # S2 — transform to trajectory groups
         groups, tmetrics = transform_episodes_to_trajectory_groups(
