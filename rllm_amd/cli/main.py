@@ -25,6 +25,7 @@ class LazyGroup(click.Group):
         "model": "rllm_amd.cli.model_cmd:model_cmd",
         "init": "rllm_amd.cli.init_cmd:init_cmd",
         "snapshot": "rllm_amd.cli.snapshot_cmd:snapshot_cmd",
+        "ui": "rllm_amd.cli.ui_cmd:ui",
     }
 
     def list_commands(self, ctx):

@@ -255,10 +255,13 @@ class UnifiedTrainer:
         metrics["batch/num_episodes"] = len(episodes)
         # episode logs carry the post-advantage training payloads (in
         # global-batch mode every rank holds the same episodes: rank 0 writes)
-        if self.episode_logger and (not (cfg.global_batch_mode and world > 1)
-                                    or pdist.get_rank() == 0):
-            self.episode_logger.log_episodes(episodes, mode="train",
-                                             step=self.state.global_step, epoch=self.state.epoch)
+        if not (cfg.global_batch_mode and world > 1) or pdist.get_rank() == 0:
+            if self.episode_logger:
+                self.episode_logger.log_episodes(episodes, mode="train",
+                                                 step=self.state.global_step,
+                                                 epoch=self.state.epoch)
+            # live streaming to UI backends (reference UILogger)
+            self.tracking.log_episodes(episodes, step=self.state.global_step)
         return metrics
 
     # ------------------------------------------------------------------

@@ -162,6 +162,12 @@ class Tracking:
                     self.loggers.append(TensorboardLogger(str(Path(log_dir) / experiment_name)))
                 elif b == "mlflow":
                     self.loggers.append(MlflowLogger(project_name, experiment_name))
+                elif b == "ui":
+                    import os
+
+                    self.loggers.append(UILogger(
+                        os.environ.get("RLLM_UI_URL", "http://127.0.0.1:3000"),
+                        run_id=experiment_name))
                 elif b.startswith("ui:"):
                     self.loggers.append(UILogger(b[3:], run_id=experiment_name))
                 else:
@@ -175,6 +181,20 @@ class Tracking:
                 lg.log(metrics, step)
             except Exception as e:  # noqa: BLE001
                 logger.warning("tracking log failed (%s): %s", type(lg).__name__, e)
+
+    def log_episodes(self, episodes, step: int):
+        """Stream full episodes to backends that accept them (UILogger live
+        view — reference tracking.py:320-487)."""
+        sinks = [lg for lg in self.loggers if hasattr(lg, "log_episode")]
+        if not sinks:
+            return
+        for ep in episodes:
+            d = ep.to_dict() if hasattr(ep, "to_dict") else ep
+            for lg in sinks:
+                try:
+                    lg.log_episode(d, step)
+                except Exception as e:  # noqa: BLE001
+                    logger.warning("episode stream failed (%s): %s", type(lg).__name__, e)
 
     def finish(self):
         for lg in self.loggers:
