@@ -148,3 +148,29 @@ def test_snapshot_without_docker(tmp_path):
     r = CliRunner().invoke(cli, ["snapshot", "create", str(out), "--no-pull"])
     assert r.exit_code == 0
     assert "snapshots ready" in r.output
+
+
+def test_dataset_from_eval_to_sft_round_trip(tmp_path):
+    """episodes -> curation -> SFT jsonl -> `sft --dry-run` consumes it."""
+    from rllm_amd.types import Episode, Step, Trajectory
+    from rllm_amd.utils.episode_logger import EpisodeLogger
+
+    eps = []
+    for i in range(4):
+        ok = i % 2 == 0
+        st = Step(chat_completions=[{"role": "user", "content": f"q{i}"},
+                                    {"role": "assistant", "content": f"a{i}"}],
+                  model_response=f"a{i}", reward=float(ok))
+        eps.append(Episode(id=f"t{i}:0", trajectories=[Trajectory(name="s", steps=[st], reward=float(ok))],
+                           is_correct=ok))
+    d = tmp_path / "episodes"
+    EpisodeLogger(str(d)).log_episodes(eps, mode="eval", step=0, epoch=0)
+
+    out = tmp_path / "sft.jsonl"
+    r = CliRunner().invoke(cli, ["dataset", "from-eval", str(d),
+                                 "--filter", "is_correct == True", "-o", str(out)])
+    assert r.exit_code == 0, r.output
+    assert "curated 2 SFT rows" in r.output
+    r = CliRunner().invoke(cli, ["sft", str(out), "--dry-run"])
+    assert r.exit_code == 0, r.output
+    assert "2 rows" in r.output
