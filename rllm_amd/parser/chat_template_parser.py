@@ -65,6 +65,8 @@ class ChatTemplateParser:
             return DeepseekQwenChatTemplateParser(tokenizer)
         if "llama" in fam:
             return LlamaChatTemplateParser(tokenizer)
+        if "vl" in fam or "vision" in fam:
+            return QwenVLChatTemplateParser(tokenizer)
         return QwenChatTemplateParser(tokenizer)
 
 
@@ -77,6 +79,53 @@ class QwenChatTemplateParser(ChatTemplateParser):
         if isinstance(content, list):  # multimodal blocks: text parts only
             content = "".join(b.get("text", "") for b in content if isinstance(b, dict))
         return f"<|im_start|>{role}\n{content}<|im_end|>\n"
+
+
+class QwenVLChatTemplateParser(QwenChatTemplateParser):
+    """Qwen2-VL family (reference chat_template_parser.py:578
+    process_image_data): multimodal content parts render vision
+    placeholders inline with the text; `extract_image_data` pulls the
+    actual image payloads for the engine/server (the gateway passes them
+    through untouched — traces carry TOKENS, images never enter the
+    training payload)."""
+
+    VISION_BLOCK = "<|vision_start|><|image_pad|><|vision_end|>"
+
+    def format_message(self, message: dict[str, Any]) -> str:
+        role = message.get("role", "user")
+        content = message.get("content") or ""
+        if isinstance(content, list):
+            parts = []
+            for b in content:
+                if not isinstance(b, dict):
+                    continue
+                t = b.get("type", "text")
+                if t == "text":
+                    parts.append(b.get("text", ""))
+                elif t in ("image", "image_url"):
+                    parts.append(self.VISION_BLOCK)
+                elif t == "video":
+                    parts.append("<|vision_start|><|video_pad|><|vision_end|>")
+            content = "".join(parts)
+        return f"<|im_start|>{role}\n{content}<|im_end|>\n"
+
+    @staticmethod
+    def extract_image_data(messages: list[dict]) -> list:
+        """Image payloads (urls / base64 data) in message order."""
+        out = []
+        for m in messages:
+            content = m.get("content")
+            if not isinstance(content, list):
+                continue
+            for b in content:
+                if not isinstance(b, dict):
+                    continue
+                if b.get("type") == "image_url":
+                    u = b.get("image_url")
+                    out.append(u.get("url") if isinstance(u, dict) else u)
+                elif b.get("type") == "image":
+                    out.append(b.get("image"))
+        return out
 
 
 class DeepseekQwenChatTemplateParser(ChatTemplateParser):

@@ -149,3 +149,31 @@ def test_qwen_parser_matches_hf_jinja_template():
         ours_nogen = parser.format(msgs, add_generation_prompt=False)
         theirs_nogen = hf_tok.apply_chat_template(msgs, tokenize=False, add_generation_prompt=False)
         assert ours_nogen == theirs_nogen
+
+
+def test_qwen_vl_parser_multimodal_blocks():
+    """Qwen-VL parser (reference chat_template_parser.py:578): image parts
+    render vision placeholders; extract_image_data pulls the payloads."""
+    from rllm_amd.parser.chat_template_parser import ChatTemplateParser, QwenVLChatTemplateParser
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    p = QwenVLChatTemplateParser(ByteTokenizer())
+    msgs = [{"role": "user", "content": [
+        {"type": "text", "text": "what is in "},
+        {"type": "image_url", "image_url": {"url": "data:image/png;base64,AAAA"}},
+        {"type": "text", "text": " this image?"},
+    ]}]
+    text = p.format(msgs, add_generation_prompt=True)
+    assert "<|vision_start|><|image_pad|><|vision_end|>" in text
+    assert "what is in " in text and " this image?" in text
+    assert text.endswith("<|im_start|>assistant\n")
+    imgs = QwenVLChatTemplateParser.extract_image_data(msgs)
+    assert imgs == ["data:image/png;base64,AAAA"]
+    # family dispatch
+    assert isinstance(ChatTemplateParser.get_parser(ByteTokenizer(), "qwen2.5-vl-7b"),
+                      QwenVLChatTemplateParser)
+    # plain-text messages unchanged vs base Qwen parser
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+
+    plain = [{"role": "user", "content": "hi"}]
+    assert p.format(plain) == QwenChatTemplateParser(ByteTokenizer()).format(plain)
