@@ -299,3 +299,36 @@ def test_alias_old_logprob_bit_identical_to_recompute():
     # recompute must report the identical value (old_lp bit-equal)
     assert m_re["offpolicy/abs_diff_max"] == m_al["offpolicy/abs_diff_max"]
     assert m_re["offpolicy/kl"] == m_al["offpolicy/kl"]
+
+
+@requires_gpu
+def test_update_is_run_to_run_deterministic():
+    """Two identical updates produce BIT-identical weights — guards the
+    deterministic grad-norm reduction (no atomics) and kernel determinism
+    across the whole fwd/bwd/optim path (sanitizer lane companion)."""
+    import random
+
+    from rllm_amd.trainer.batch import PackedRow
+
+    def run():
+        torch.manual_seed(4)
+        model = tiny_model(seed=33)
+        ref = tiny_model(seed=33)
+        for p in ref.parameters():
+            p.requires_grad_(False)
+        t = PolicyTrainer(model, ref, PolicyTrainerConfig(
+            lr=1e-3, kl_beta=1e-2, grad_clip=1.0, old_logprob_mode="alias",
+            max_tokens_per_micro=128))
+        rng = random.Random(11)
+        rows = []
+        for _ in range(5):
+            n = rng.randint(24, 48)
+            rows.append(PackedRow(tokens=[rng.randrange(1024) for _ in range(n)],
+                                  response_mask=[0] * 4 + [1] * (n - 4),
+                                  advantages=[rng.uniform(-1, 1)] * n,
+                                  rollout_logprobs=[-1.0] * n))
+        t.update_policy(rows)
+        return t.flat_param.clone()
+
+    a, b = run(), run()
+    assert torch.equal(a, b), (a - b).abs().max()
