@@ -541,7 +541,9 @@ class LLMEngine:
             # token/logprob history read back ONCE per block
             "slot_sched": torch.zeros(K, Bmax, device=dev, dtype=torch.int32),
             "blk_i": torch.zeros(1, device=dev, dtype=torch.long),
-            "tok_hist": torch.zeros(K, Bmax, device=dev, dtype=torch.long),
+            # int32: the fused sampler emits int32 tokens and index_copy_
+            # requires matching dtypes
+            "tok_hist": torch.zeros(K, Bmax, device=dev, dtype=torch.int32),
             "lp_hist": torch.zeros(K, Bmax, device=dev, dtype=torch.float32),
         }
         self._hb = {
@@ -612,7 +614,18 @@ class LLMEngine:
             R = min(R,
                     s.params.max_tokens - len(s.output_ids),
                     self.max_model_len - s.total_len)
-        return max(1, R)
+        R = max(1, R)
+        # shrink until the whole block's page reservation fits the pool —
+        # a tight pool must behave like single-step decode (preemption
+        # logic unchanged), not livelock on an oversized reservation
+        free = self._effective_free_pages()
+        while R > 1:
+            need = sum(max(0, KVCache.pages_needed(s.total_len + R) - len(s.pages))
+                       for s in batch)
+            if need <= free:
+                break
+            R //= 2
+        return R
 
     def _run_decode_graph(self, batch: list[Sequence], temp: float) -> int:
         if self._gb is None:

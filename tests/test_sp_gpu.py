@@ -81,11 +81,18 @@ def test_sp_update_matches_single_process():
     torch.cuda.empty_cache()
 
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     procs = [ctx.Process(target=_sp_worker, args=(r, 2, q)) for r in range(2)]
     for p in procs:
         p.start()
-    results = [q.get(), q.get()]
+    results = []
+    try:
+        for _ in range(2):
+            results.append(q.get(timeout=420))
+    except Exception:
+        for p in procs:
+            p.terminate()
+        pytest.fail("SP worker died or hung (no result within 420s)")
     for p in procs:
         p.join(timeout=120)
         if p.is_alive():
