@@ -125,11 +125,15 @@ _SYMPY_POOL = None
 def _get_sympy_pool():
     global _SYMPY_POOL
     if _SYMPY_POOL is None:
+        import atexit
         import concurrent.futures as cf
         import multiprocessing as mp
 
         _SYMPY_POOL = cf.ProcessPoolExecutor(
             max_workers=1, mp_context=mp.get_context("spawn"))
+        # tear down explicitly — a live pool at interpreter exit trips
+        # noisy weakref callbacks in concurrent.futures
+        atexit.register(lambda: _SYMPY_POOL and _SYMPY_POOL.shutdown(wait=False, cancel_futures=True))
     return _SYMPY_POOL
 
 
