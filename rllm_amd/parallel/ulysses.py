@@ -32,8 +32,16 @@ def sp_world(group=None) -> int:
 def _a2a(x: torch.Tensor, group) -> torch.Tensor:
     """all_to_all_single over the leading [P, ...] axis (equal splits).
     Both buffers MUST be dense: the collective reads/writes flat memory, so
-    a strided empty_like of a permuted view would be silently garbled."""
+    a strided empty_like of a permuted view would be silently garbled.
+
+    gloo has no CUDA all-to-all — stage through host there (test/fallback
+    path only; RCCL runs it natively over xGMI)."""
     x = x.contiguous()
+    if x.is_cuda and dist.get_backend(group) == "gloo":
+        xc = x.cpu()
+        out = torch.empty_like(xc)
+        dist.all_to_all_single(out, xc, group=group)
+        return out.to(x.device)
     out = torch.empty_like(x)
     dist.all_to_all_single(out, x, group=group)
     return out

@@ -61,7 +61,9 @@ def _sp_worker(rank, world, q):
         torch.cuda.set_device(0)
         trainer = _build_trainer(sp=True)
         m = trainer.update_policy(_rows())
-        flat = trainer.flat_param.float().cpu()
+        # numpy copy: torch tensors ride shared memory through mp queues,
+        # which breaks once the producer process exits
+        flat = trainer.flat_param.float().cpu().numpy().copy()
         q.put((rank, "ok", m["actor/loss"], m["actor/grad_norm"], flat))
         pdist.destroy()
     except Exception as e:  # noqa: BLE001
@@ -103,6 +105,7 @@ def test_sp_update_matches_single_process():
 
     _, _, loss0, gnorm0, flat0 = results[0]
     _, _, loss1, gnorm1, flat1 = results[1]
+    flat0, flat1 = torch.from_numpy(flat0), torch.from_numpy(flat1)
     # both SP ranks end bit-identical (same all-reduced grads)
     assert torch.equal(flat0, flat1)
     # and match the single-process update within bf16/a2a reorder noise
