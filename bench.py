@@ -46,6 +46,8 @@ def parse_args():
     p.add_argument("--max-new-tokens", type=int, default=512, help="response length cap per rollout")
     p.add_argument("--micro-tokens", type=int, default=0,
                    help="ppo_max_token_len_per_gpu (0 = auto by model size)")
+    p.add_argument("--max-batched-tokens", type=int, default=8192,
+                   help="prefill chunk size (engine knob, not an algorithmic config)")
     p.add_argument("--kl-beta", type=float, default=1e-3)
     p.add_argument("--lr", type=float, default=1e-6)
     p.add_argument("--old-logprob-mode", default="alias",
@@ -117,7 +119,7 @@ def main():
     # ranks sharing one device (gloo smoke on a 1-GPU box) split the budget
     ranks_per_dev = max(1, (world + n_devices - 1) // n_devices)
     engine = LLMEngine(
-        model, max_num_seqs=1024, max_num_batched_tokens=8192,
+        model, max_num_seqs=1024, max_num_batched_tokens=args.max_batched_tokens,
         kv_budget_bytes=min(int(free * 0.35) // ranks_per_dev, 64 << 30),
         eos_token_id=None,  # synthetic data: length-capped rollouts
         seed=args.seed * 1000 + rank)
