@@ -305,3 +305,31 @@ HARNESS_REGISTRY.update({
     "opencode": OpenCodeHarness,
     "qwen-code": QwenCodeHarness,
 })
+
+
+class ReactHarness(BaseCliHarness):
+    """One-shot LLM call through the gateway — the default scaffold for
+    data tasks (math/MCQ/QA; reference harnesses/react.py). Pure python:
+    no sandbox binary, no install; works with needs_env=False."""
+
+    name = "react"
+    needs_env = False
+    system_prompt = ("Solve the task. Think step by step, then give the final "
+                     "answer in \\boxed{}.")
+
+    async def arun(self, task: Task, config: AgentConfig, *, env: Sandbox | None = None) -> None:
+        import httpx
+
+        async with httpx.AsyncClient(timeout=self.timeout) as client:
+            r = await client.post(
+                config.base_url + "/chat/completions",
+                json={"model": config.model,
+                      "messages": [
+                          {"role": "system", "content": self.system_prompt},
+                          {"role": "user", "content": str(task.instruction)},
+                      ]})
+            r.raise_for_status()
+        return None  # Episode assembled from gateway traces
+
+
+HARNESS_REGISTRY["react"] = ReactHarness

@@ -195,3 +195,45 @@ def test_harness_e2e_episode_from_traces():
         assert st.response_ids and st.logprobs
         assert len(st.response_ids) == len(st.logprobs)
     assert ep.is_correct and traj.reward == 1.0
+
+
+def test_react_harness_one_shot():
+    """react scaffold: one chat call through the real gateway, episode
+    from traces, no sandbox needed."""
+    from helpers.fake_worker import FakeWorkerServer
+
+    import rllm_amd
+
+    @rllm_amd.evaluator
+    def ev(task, episode):
+        return 1.0
+
+    with FakeWorkerServer() as worker:
+        from rllm_amd.gateway.manager import GatewayManager
+        from rllm_amd.gateway.models import GatewayConfig
+
+        gw = GatewayManager(GatewayConfig())
+        gw.start(worker_urls=[worker.url])
+        try:
+            engine = AgentFlowEngine(H.get_harness("react"), gw, evaluator=ev,
+                                     n_parallel_tasks=2)
+            eps = asyncio.run(engine.execute_tasks(
+                [Task(id="r1", instruction="2+2?")], ["r1:0"]))
+        finally:
+            gw.stop()
+    assert len(eps) == 1 and eps[0].trajectories[0].steps[0].response_ids
+
+
+def test_agent_registry_resolves():
+    import importlib
+    import json
+
+    reg = json.loads((Path("rllm_amd/registry/agents.json")).read_text())
+    n = 0
+    for name, e in reg.items():
+        if not isinstance(e, dict) or "module" not in e:
+            continue
+        cls = getattr(importlib.import_module(e["module"]), e["function"])
+        assert callable(cls), name
+        n += 1
+    assert n >= 10
