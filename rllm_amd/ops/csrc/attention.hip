@@ -586,8 +586,11 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor k_pages, torch::Tensor
   auto o = torch::empty_like(q);
 
   if (n_splits <= 0) {
-    // heuristic: ~4 blocks per CU for latency hiding, capped at 16
-    int target = (4 * 256) / std::max(1, B * Hk);
+    // measured (profiles/paged_sweep.log, B=256 Hk=2, S=256..8192): once
+    // B*Hk blocks already cover the 256 CUs ~2x, splitting only adds the
+    // fp32 workspace round-trip + merge kernel (-5us/call). Split ONLY to
+    // fill the chip at small batch, capped at 16.
+    int target = (2 * 256) / std::max(1, B * Hk);
     n_splits = std::min(16, std::max(1, target));
   }
   dim3 grid(B, Hk, (unsigned)n_splits);
