@@ -11,9 +11,12 @@ from rllm_amd.harnesses.coding_agents import (
     AiderHarness,
     ClaudeCodeHarness,
     CodexHarness,
+    KimiCliHarness,
     MiniSweAgentHarness,
     OpenCodeHarness,
     QwenCodeHarness,
+    ReactHarness,
+    ZeroClawHarness,
 )
 
 __all__ = [
@@ -21,4 +24,5 @@ __all__ = [
     "BashHarness", "OracleHarness", "CurlChatHarness",
     "MiniSweAgentHarness", "AiderHarness", "ClaudeCodeHarness",
     "CodexHarness", "OpenCodeHarness", "QwenCodeHarness",
+    "KimiCliHarness", "ZeroClawHarness", "ReactHarness",
 ]
