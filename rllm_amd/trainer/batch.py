@@ -164,58 +164,68 @@ def pack_rows(rows: list[PackedRow], device: str = "cpu", max_seq_len: int | Non
     pad_to_multiple: append a dummy fully-masked sequence so the packed
     length divides the Ulysses SP degree (parallel/ulysses.py shard_slice).
     """
-    input_ids: list[int] = []
-    positions: list[int] = []
+    import numpy as np
+
+    ids_parts: list[np.ndarray] = []
+    pos_parts: list[np.ndarray] = []
+    lm_parts: list[np.ndarray] = []
+    tgt_parts: list[np.ndarray] = []
+    adv_parts: list[np.ndarray] = []
+    rl_parts: list[np.ndarray] = []
     cu = [0]
-    loss_mask: list[bool] = []
-    targets: list[int] = []
-    advantages: list[float] = []
-    rollout_lps: list[float] = []
     n_resp = 0
 
     for row in rows:
-        toks = row.tokens
-        if max_seq_len is not None and len(toks) > max_seq_len:
-            toks = toks[:max_seq_len]
-        n = len(toks)
+        n = len(row.tokens)
+        if max_seq_len is not None and n > max_seq_len:
+            n = max_seq_len
         if n < 2:
             continue
-        input_ids.extend(toks)
-        positions.extend(range(n))
-        cu.append(cu[-1] + n)
-        for r in range(n):
-            if r + 1 < n and row.response_mask[r + 1]:
-                loss_mask.append(True)
-                targets.append(row.tokens[r + 1])
-                advantages.append(row.advantages[r + 1])
-                rollout_lps.append(row.rollout_logprobs[r + 1])
-                n_resp += 1
-            else:
-                loss_mask.append(False)
-                targets.append(0)
-                advantages.append(0.0)
-                rollout_lps.append(0.0)
+        toks = np.asarray(row.tokens[:n], dtype=np.int64)
+        mask = np.asarray(row.response_mask[:n], dtype=bool)
+        # shift semantics: row r predicts token r+1 — a row is a loss row
+        # when the NEXT token is an action token; the last row never is
+        lm = np.empty(n, dtype=bool)
+        lm[:-1] = mask[1:]
+        lm[-1] = False
+        tgt = np.zeros(n, dtype=np.int64)
+        tgt[:-1] = np.where(mask[1:], toks[1:], 0)
+        adv = np.zeros(n, dtype=np.float32)
+        adv[:-1] = np.where(mask[1:], np.asarray(row.advantages[1:n], dtype=np.float32), 0.0)
+        rl = np.zeros(n, dtype=np.float32)
+        rl[:-1] = np.where(mask[1:], np.asarray(row.rollout_logprobs[1:n], dtype=np.float32), 0.0)
 
-    if pad_to_multiple and len(input_ids) % pad_to_multiple:
-        pad = pad_to_multiple - len(input_ids) % pad_to_multiple
+        ids_parts.append(toks)
+        pos_parts.append(np.arange(n, dtype=np.int32))
+        lm_parts.append(lm)
+        tgt_parts.append(tgt)
+        adv_parts.append(adv)
+        rl_parts.append(rl)
+        cu.append(cu[-1] + n)
+        n_resp += int(lm.sum())
+
+    total = cu[-1]
+    if pad_to_multiple and total % pad_to_multiple:
+        pad = pad_to_multiple - total % pad_to_multiple
         if pad == 1:  # a 1-token sequence is degenerate; pad a full extra round
             pad += pad_to_multiple
-        input_ids.extend([0] * pad)
-        positions.extend(range(pad))
+        ids_parts.append(np.zeros(pad, dtype=np.int64))
+        pos_parts.append(np.arange(pad, dtype=np.int32))
+        lm_parts.append(np.zeros(pad, dtype=bool))
+        tgt_parts.append(np.zeros(pad, dtype=np.int64))
+        adv_parts.append(np.zeros(pad, dtype=np.float32))
+        rl_parts.append(np.zeros(pad, dtype=np.float32))
         cu.append(cu[-1] + pad)
-        loss_mask.extend([False] * pad)
-        targets.extend([0] * pad)
-        advantages.extend([0.0] * pad)
-        rollout_lps.extend([0.0] * pad)
 
+    cat = (lambda parts, dt: np.concatenate(parts) if parts else np.empty(0, dtype=dt))
     return TrainBatch(
-        input_ids=torch.tensor(input_ids, dtype=torch.long, device=device),
-        positions=torch.tensor(positions, dtype=torch.int32, device=device),
+        input_ids=torch.from_numpy(cat(ids_parts, np.int64)).to(device),
+        positions=torch.from_numpy(cat(pos_parts, np.int32)).to(device),
         cu_seqlens=cu,
-        loss_mask=torch.tensor(loss_mask, dtype=torch.bool, device=device),
-        targets=torch.tensor(targets, dtype=torch.long, device=device),
-        advantages=torch.tensor(advantages, dtype=torch.float32, device=device),
-        rollout_logprobs=torch.tensor(rollout_lps, dtype=torch.float32, device=device),
+        loss_mask=torch.from_numpy(cat(lm_parts, bool)).to(device),
+        targets=torch.from_numpy(cat(tgt_parts, np.int64)).to(device),
+        advantages=torch.from_numpy(cat(adv_parts, np.float32)).to(device),
+        rollout_logprobs=torch.from_numpy(cat(rl_parts, np.float32)).to(device),
         n_response_tokens=n_resp,
         n_rows=len(cu) - 1,
     )
