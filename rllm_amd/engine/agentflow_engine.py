@@ -125,7 +125,11 @@ class AgentFlowEngine:
         self.rollout_timeout = rollout_timeout
         self.raise_on_error = raise_on_error
         self.default_sampling_params = default_sampling_params or {}
-        self.executor = ThreadPoolExecutor(max_workers=min(n_parallel_tasks, 64))
+        # one thread per parallel task: flows block in sync HTTP for the
+        # WHOLE rollout, so capping threads below the semaphore silently
+        # caps the engine's decode batch (measured 4x tok/s loss at 256
+        # sessions with a 64-thread cap — profiles/flowpath.log)
+        self.executor = ThreadPoolExecutor(max_workers=min(n_parallel_tasks, 512))
         self._accepts_env = flow_accepts_env(agent_flow)
         self.is_validation = False
 
