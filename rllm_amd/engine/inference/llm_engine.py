@@ -142,6 +142,9 @@ class LLMEngine:
         self.running: list[Sequence] = []
         self.finished: dict[str, Sequence] = {}
         self._sample_step = 0
+        # batching diagnostics (perf scripts read these)
+        self.stats = {"decode_calls": 0, "decode_batch_sum": 0, "decode_tokens": 0,
+                      "prefill_calls": 0, "prefill_tokens": 0}
 
         # ---- hipGraph decode state (launch-bound decode: ~300 dispatches /
         # token collapse into one graph replay; guide §1 "capture
@@ -294,6 +297,8 @@ class LLMEngine:
             n_tok = self._run_prefill_cached(batch)
         else:
             n_tok = self._run_prefill_full(batch)
+        self.stats["prefill_calls"] += 1
+        self.stats["prefill_tokens"] += n_tok
         for seq in batch:
             if seq.state != SeqState.FINISHED:
                 seq.state = SeqState.RUNNING
@@ -479,6 +484,9 @@ class LLMEngine:
             n = self._run_decode_graph(batch, temp)
         else:
             n = self._run_decode_eager(batch)
+        self.stats["decode_calls"] += 1
+        self.stats["decode_batch_sum"] += len(batch)
+        self.stats["decode_tokens"] += n
         self.running = [s for s in batch if s.state == SeqState.RUNNING]
         return n
 

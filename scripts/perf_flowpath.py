@@ -84,6 +84,15 @@ def main():
         m["time/step_total_s"] = time.monotonic() - t0
         times.append({k: v for k, v in m.items() if k.startswith("time/") or k.startswith("batch/")})
         print(f"step: { {k: round(v, 2) for k, v in times[-1].items()} }", flush=True)
+        st = backend.engine.stats
+        if st["decode_calls"]:
+            print(f"engine: decode_calls={st['decode_calls']} "
+                  f"avg_decode_batch={st['decode_batch_sum']/st['decode_calls']:.1f} "
+                  f"decode_tokens={st['decode_tokens']} "
+                  f"prefill_calls={st['prefill_calls']} prefill_tokens={st['prefill_tokens']}",
+                  flush=True)
+            for k in st:
+                st[k] = 0
         return m
 
     trainer._train_batch_async = timed
