@@ -130,6 +130,14 @@ class AgentFlowEngine:
         # caps the engine's decode batch (measured 4x tok/s loss at 256
         # sessions with a 64-thread cap — profiles/flowpath.log)
         self.executor = ThreadPoolExecutor(max_workers=min(n_parallel_tasks, 512))
+        # shared pooled HTTP client handed to flows via config.http (per-call
+        # httpx.Client construction is GIL-serialized and throttles arrival)
+        import httpx as _httpx
+
+        self.http = _httpx.Client(
+            timeout=600.0,
+            limits=_httpx.Limits(max_connections=max(n_parallel_tasks * 2, 64),
+                                 max_keepalive_connections=max(n_parallel_tasks * 2, 64)))
         self._accepts_env = flow_accepts_env(agent_flow)
         self.is_validation = False
 
@@ -190,7 +198,8 @@ class AgentFlowEngine:
             session_url = await self.gateway.acreate_session(uid, self.default_sampling_params)
             config = AgentConfig(base_url=session_url, model=self.model_name, session_uid=uid,
                                  is_validation=self.is_validation,
-                                 sampling_params=dict(self.default_sampling_params))
+                                 sampling_params=dict(self.default_sampling_params),
+                                 http=self.http)
             episode = await asyncio.wait_for(
                 run_agent_flow(self.agent_flow, task, config, executor=self.executor,
                                env=ctx.env if self._accepts_env else None),

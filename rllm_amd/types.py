@@ -518,6 +518,22 @@ class AgentConfig:
     metadata: dict = field(default_factory=dict)
     is_validation: bool = False
     sampling_params: dict = field(default_factory=dict)
+    # shared httpx.Client (thread-safe, pooled) supplied by the engine.
+    # Flows SHOULD use `config.http.post(...)` instead of building a client
+    # per call: at 256 parallel rollouts, per-call client construction is
+    # GIL-serialized (~30 ms each) and alone throttled request arrival to
+    # ~25/s, starving the decode batch. None when running outside an engine.
+    http: Any = None
+
+    def post(self, path: str, json: dict, timeout: float = 600.0):
+        """Convenience: POST {base_url}{path} with the shared client (or a
+        one-shot httpx call outside an engine)."""
+        url = self.base_url.rstrip("/") + path
+        if self.http is not None:
+            return self.http.post(url, json=json, timeout=timeout)
+        import httpx
+
+        return httpx.post(url, json=json, timeout=timeout)
 
 
 @runtime_checkable

@@ -41,10 +41,11 @@ def main():
 
     @rllm_amd.rollout
     def flow(task, config):
-        r = httpx.post(config.base_url + "/chat/completions",
-                       json={"model": config.model,
-                             "messages": [{"role": "user", "content": str(task.instruction)}]},
-                       timeout=600.0)
+        # config.post uses the engine's SHARED pooled client — essential at
+        # this concurrency (per-call clients throttle arrival to ~25/s)
+        r = config.post("/chat/completions",
+                        json={"model": config.model,
+                              "messages": [{"role": "user", "content": str(task.instruction)}]})
         r.raise_for_status()
         return None
 
