@@ -64,6 +64,9 @@ class FullyAsyncTrainer(UnifiedTrainer):
                         tasks, uids = interleave_tasks([task], cfg.rollout_n)
                         try:
                             episodes = await self.backend.generate_episodes(tasks, uids=uids)
+                            # S1.5 (distillation teacher fetch etc.) — same
+                            # stage the synchronous loop runs
+                            self.backend.postprocess_episodes(episodes)
                             for ep in episodes:
                                 await buffer.add_episode(ep)
                         finally:
