@@ -57,6 +57,7 @@ class NativeBackend(BackendProtocol):
         max_prompt_length: int | None = None,
         rollout_sampling_params: dict | None = None,
         n_parallel_tasks: int = 128,
+        hooks=None,  # TaskHooks (e.g. SandboxTaskHooks for sandboxed agents)
         checkpoint_path: str | None = None,
         seed: int = 0,
         lora=None,  # models.lora.LoRAConfig | True for defaults | None = full finetune
@@ -101,6 +102,7 @@ class NativeBackend(BackendProtocol):
         # (reference verl_backend.py:612 meta_info['temperature'])
         self.policy_config.temperature = float(self.rollout_sampling_params.get("temperature", 1.0))
         self.n_parallel_tasks = n_parallel_tasks
+        self.hooks = hooks
         self.checkpoint_path = checkpoint_path
         self.lora_config = lora
         self.gateway_config = gateway_config
@@ -177,7 +179,8 @@ class NativeBackend(BackendProtocol):
         self.gateway.start()
         self.flow_engine = AgentFlowEngine(
             self.agent_flow, self.gateway, model_name=self.cfg.name,
-            evaluator=self.evaluator, n_parallel_tasks=self.n_parallel_tasks,
+            evaluator=self.evaluator, hooks=self.hooks,
+            n_parallel_tasks=self.n_parallel_tasks,
             default_sampling_params=self.rollout_sampling_params)
         return self.rollout_engine
 
