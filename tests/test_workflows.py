@@ -148,3 +148,60 @@ def test_distillation_workflow_attaches_advantages():
     step = ep.trajectories[0].steps[0]
     # student logprobs -0.1 each, teacher -0.05 -> adv = 2*(0.05) = 0.1
     assert step.advantage == pytest.approx([0.1, 0.1])
+
+
+def test_tool_calling_flow_e2e():
+    """ToolCallingFlow (reference ToolCallingMixin): model emits a
+    <tool_call>, the loop executes it and feeds the result back; episode
+    assembled from the gateway traces has one step per LLM turn."""
+    import asyncio
+    import json as _json
+
+    from rllm_amd.engine.agentflow_engine import AgentFlowEngine
+    from rllm_amd.gateway.manager import GatewayManager
+    from rllm_amd.gateway.models import GatewayConfig
+    from rllm_amd.tools.python_tool import CalculatorTool
+    from rllm_amd.types import Task
+    from rllm_amd.workflows.tool_loop import ToolCallingFlow
+
+    turn_log = []
+
+    async def handler(request):
+        body = request["body"]
+        msgs = body["messages"]
+        turn_log.append(len(msgs))
+        if not any(m["role"] == "tool" for m in msgs):
+            content = ('I will compute it.\n<tool_call>\n'
+                       '{"name": "calculator", "arguments": {"expression": "6*7"}}\n'
+                       '</tool_call>')
+        else:
+            tool_msg = next(m for m in msgs if m["role"] == "tool")
+            value = _json.loads(tool_msg["content"])["output"]
+            content = f"The answer is {value}."
+        n = 4
+        return {"id": "x", "object": "chat.completion", "model": "m",
+                "choices": [{"index": 0,
+                             "message": {"role": "assistant", "content": content},
+                             "token_ids": list(range(n)), "prompt_token_ids": [1, 2, 3],
+                             "logprobs": {"token_logprobs": [-0.1] * n},
+                             "finish_reason": "stop"}],
+                "usage": {}}
+
+    gw = GatewayManager(GatewayConfig(), local_handler=handler)
+    gw.start()
+    try:
+        flow = ToolCallingFlow([CalculatorTool()], max_turns=4)
+        engine = AgentFlowEngine(flow, gw, evaluator=None, n_parallel_tasks=2)
+        eps = asyncio.run(engine.execute_tasks(
+            [Task(id="t", instruction="what is 6*7?")], ["t:0"]))
+    finally:
+        gw.stop()
+
+    assert len(turn_log) == 2  # tool turn + answer turn
+    ep = eps[0]
+    steps = ep.trajectories[0].steps
+    assert len(steps) == 2
+    assert "tool_call" in steps[0].model_response
+    assert "The answer is 42" in steps[1].model_response
+    for st in steps:
+        assert st.response_ids and len(st.logprobs) == len(st.response_ids)
