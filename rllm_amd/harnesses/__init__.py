@@ -7,6 +7,7 @@ from rllm_amd.harnesses.cli_harness import (
     OracleHarness,
     get_harness,
 )
+from rllm_amd.harnesses.terminus2 import Terminus2Harness
 from rllm_amd.harnesses.coding_agents import (
     AiderHarness,
     ClaudeCodeHarness,
@@ -24,5 +25,5 @@ __all__ = [
     "BashHarness", "OracleHarness", "CurlChatHarness",
     "MiniSweAgentHarness", "AiderHarness", "ClaudeCodeHarness",
     "CodexHarness", "OpenCodeHarness", "QwenCodeHarness",
-    "KimiCliHarness", "ZeroClawHarness", "ReactHarness",
+    "KimiCliHarness", "ZeroClawHarness", "ReactHarness", "Terminus2Harness",
 ]
