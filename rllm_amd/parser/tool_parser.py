@@ -14,6 +14,11 @@ class ToolParser:
     def parse(self, text: str) -> list[ToolCall]:
         raise NotImplementedError
 
+    def get_tool_prompt(self, tools_schema: str) -> str:
+        """System-prompt fragment teaching the model this parser's wire
+        format (reference tool_parser.py get_tool_prompt)."""
+        raise NotImplementedError
+
     @staticmethod
     def get_parser(model_family: str = "qwen") -> "ToolParser":
         fam = (model_family or "qwen").lower()
@@ -42,15 +47,65 @@ class QwenToolParser(ToolParser):
     def format_call(call: ToolCall) -> str:
         return f"<tool_call>\n{json.dumps({'name': call.name, 'arguments': call.arguments})}\n</tool_call>"
 
+    def get_tool_prompt(self, tools_schema: str) -> str:
+        return (
+            "\n# Tools\n\nYou may call one or more functions to assist with "
+            f"the user query.\n<tools>\n{tools_schema}\n</tools>\n\n"
+            "For each function call, return a json object with function name "
+            "and arguments within <tool_call></tool_call> XML tags:\n"
+            '<tool_call>\n{"name": "<function-name>", "arguments": <args-json-object>}\n'
+            "</tool_call>\n")
+
 
 class R1ToolParser(ToolParser):
-    """R1-style: fenced ```json blocks holding {"name": ..., "arguments": ...}."""
+    """DeepSeek-R1 special-token convention:
 
-    PATTERN = re.compile(r"```json\s*\n(.*?)```", re.DOTALL)
+    <｜tool▁calls▁begin｜>
+    <｜tool▁call▁begin｜>function<｜tool▁sep｜>NAME
+    ```json
+    {args}
+    ```
+    <｜tool▁call▁end｜> ... <｜tool▁calls▁end｜>
+
+    Plain fenced ```json {"name":..., "arguments":...}``` blocks are also
+    accepted (distilled models often emit those without the specials)."""
+
+    CALL_BEGIN = "<｜tool▁call▁begin｜>"
+    CALL_END = "<｜tool▁call▁end｜>"
+    SEP = "<｜tool▁sep｜>"
+    FENCE = re.compile(r"```json\s*\n(.*?)```", re.DOTALL)
 
     def parse(self, text: str) -> list[ToolCall]:
-        calls = []
-        for m in self.PATTERN.finditer(text or ""):
+        text = text or ""
+        calls: list[ToolCall] = []
+        idx = 0
+        while True:
+            start = text.find(self.CALL_BEGIN, idx)
+            if start == -1:
+                break
+            start += len(self.CALL_BEGIN)
+            end = text.find(self.CALL_END, start)
+            if end == -1:
+                break
+            idx = end + len(self.CALL_END)
+            block = text[start:end]
+            head, _, rest = block.partition(self.SEP)
+            if not rest:
+                continue
+            name = rest.split("\n", 1)[0].strip().strip("`")
+            m = self.FENCE.search(block)
+            args: dict = {}
+            if m:
+                try:
+                    args = json.loads(m.group(1))
+                except json.JSONDecodeError:
+                    args = {"_raw": m.group(1).strip()}
+            if name:
+                calls.append(ToolCall(name=name, arguments=args))
+        if calls:
+            return calls
+        # fallback: bare fenced json with name/arguments keys
+        for m in self.FENCE.finditer(text):
             try:
                 data = json.loads(m.group(1))
             except json.JSONDecodeError:
@@ -60,3 +115,14 @@ class R1ToolParser(ToolParser):
                 if isinstance(item, dict) and "name" in item:
                     calls.append(ToolCall(name=item["name"], arguments=item.get("arguments", {})))
         return calls
+
+    def get_tool_prompt(self, tools_schema: str) -> str:
+        return (
+            "\n# Tools\n\nYou may call one or more functions to assist with "
+            f"the user query.\n<tools>\n{tools_schema}\n</tools>\n\n"
+            "Output format for tool calls:\n\n"
+            "<｜tool▁calls▁begin｜>\n"
+            "<｜tool▁call▁begin｜>function<｜tool▁sep｜>function_name\n"
+            "```json\n{\"param\": \"value\"}\n```\n"
+            "<｜tool▁call▁end｜>\n"
+            "<｜tool▁calls▁end｜>\n")

@@ -177,3 +177,33 @@ def test_qwen_vl_parser_multimodal_blocks():
 
     plain = [{"role": "user", "content": "hi"}]
     assert p.format(plain) == QwenChatTemplateParser(ByteTokenizer()).format(plain)
+
+
+def test_r1_tool_parser_special_tokens():
+    """DeepSeek-R1 special-token tool-call wire format (reference
+    tool_parser.py:47-186) plus the bare-fence fallback."""
+    from rllm_amd.parser.tool_parser import R1ToolParser
+
+    p = R1ToolParser()
+    text = ("thinking...\n<｜tool▁calls▁begin｜>\n"
+            "<｜tool▁call▁begin｜>function<｜tool▁sep｜>get_weather\n"
+            "```json\n{\"city\": \"Paris\"}\n```\n"
+            "<｜tool▁call▁end｜>\n"
+            "<｜tool▁call▁begin｜>function<｜tool▁sep｜>get_time\n"
+            "```json\n{\"tz\": \"UTC\"}\n```\n"
+            "<｜tool▁call▁end｜>\n<｜tool▁calls▁end｜>")
+    calls = p.parse(text)
+    assert [c.name for c in calls] == ["get_weather", "get_time"]
+    assert calls[0].arguments == {"city": "Paris"}
+    # fallback: plain fenced json
+    calls = p.parse('```json\n{"name": "f", "arguments": {"x": 1}}\n```')
+    assert calls[0].name == "f" and calls[0].arguments == {"x": 1}
+    # malformed args survive as _raw
+    calls = p.parse("<｜tool▁calls▁begin｜><｜tool▁call▁begin｜>function<｜tool▁sep｜>g\n"
+                    "```json\nnot json\n```\n<｜tool▁call▁end｜><｜tool▁calls▁end｜>")
+    assert calls[0].name == "g" and "_raw" in calls[0].arguments
+    # tool prompts teach the right wire format
+    assert "tool▁calls▁begin" in p.get_tool_prompt("{}")
+    from rllm_amd.parser.tool_parser import QwenToolParser
+
+    assert "<tool_call>" in QwenToolParser().get_tool_prompt("{}")
