@@ -57,10 +57,6 @@ class TrainBatch:
             setattr(self, f, getattr(self, f).to(device))
         return self
 
-    def index_select_rows(self, row_slice: slice, cu_offset: int) -> "TrainBatch":
-        raise NotImplementedError
-
-
 def _merge_cumulative(traj: Trajectory) -> PackedRow | None:
     """Merge a cumulative trajectory's steps into one packed row
     (reference transform.py:316-404 segment logic)."""
