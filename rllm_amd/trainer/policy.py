@@ -338,6 +338,9 @@ class PolicyTrainer:
                              "'alias' or 'rollout' (no recompute callback)")
         if cfg.loss_agg_mode != "token-mean":
             raise ValueError("sequence_parallel update supports token-mean only")
+        if cfg.use_ref and cfg.ref_from_lora_base:
+            raise ValueError("sequence_parallel + LoRA-base KL is not wired; "
+                             "pass an explicit ref_model or disable use_ref")
         P = pdist.get_world_size()
         rank = pdist.get_rank()
         group = self.model.sp_group or tdist.group.WORLD
