@@ -32,8 +32,8 @@ def eval_cmd(dataset, agent_ref, evaluator_ref, base_url, model, split, attempts
     tasks = registry.load_dataset(dataset, split=split, as_tasks=True)
     if limit:
         tasks = tasks[:limit]
-    flow = resolve_ref(agent_ref)
-    evaluator = resolve_ref(evaluator_ref) if evaluator_ref else None
+    flow = _resolve_agent(agent_ref, resolve_ref)
+    evaluator = _resolve_evaluator(evaluator_ref, resolve_ref)
 
     if base_url:
         gw = EvalGatewayManager(base_url)
@@ -47,3 +47,39 @@ def eval_cmd(dataset, agent_ref, evaluator_ref, base_url, model, split, attempts
         click.echo(json.dumps(result.metrics, indent=2))
     finally:
         gw.stop()
+
+
+def _resolve_agent(ref: str, resolve_ref):
+    """'module.path:attr', or a scaffold name from registry/agents.json
+    (reference eval/agent_loader.py: registry-or-module resolution)."""
+    if ":" in ref:
+        obj = resolve_ref(ref)
+        return obj() if isinstance(obj, type) else obj
+    import importlib
+    import json as _json
+    from pathlib import Path
+
+    reg = _json.loads((Path(__file__).resolve().parents[1] / "registry" / "agents.json").read_text())
+    entry = reg.get(ref)
+    if not isinstance(entry, dict) or "module" not in entry:
+        raise click.ClickException(
+            f"--agent {ref!r}: not module:attr and not a registered scaffold "
+            f"({sorted(k for k, v in reg.items() if isinstance(v, dict) and 'module' in v)})")
+    cls = getattr(importlib.import_module(entry["module"]), entry["function"])
+    return cls()
+
+
+def _resolve_evaluator(ref: str | None, resolve_ref):
+    """'module.path:attr', or a grader name from the reward registry."""
+    if ref is None:
+        return None
+    if ":" in ref:
+        return resolve_ref(ref)
+    from rllm_amd.rewards.reward_fns import REWARD_FN_REGISTRY, get_reward_fn
+
+    if ref in REWARD_FN_REGISTRY:
+        import rllm_amd
+
+        return rllm_amd.evaluator(get_reward_fn(ref))
+    raise click.ClickException(f"--evaluator {ref!r}: not module:attr and not one of "
+                               f"{sorted(REWARD_FN_REGISTRY)}")
