@@ -33,12 +33,29 @@ class StatefulTaskDataLoader:
         if self.shuffle:
             random.Random(self.seed + self.epoch).shuffle(order)
         if self.world_size > 1:
-            order = order[self.rank :: self.world_size]
+            shard = order[self.rank :: self.world_size]
+            # pad by wrapping so EVERY rank sees the same number of items —
+            # uneven shards would diverge the per-rank step count and
+            # deadlock the DP collectives on the last batch of an epoch
+            # (DistributedSampler semantics)
+            target = self._items_per_rank()
+            i = 0
+            while len(shard) < target:
+                shard.append(order[(self.rank + i) % len(order)] if order else 0)
+                i += 1
+            order = shard
         return order
 
+    def _items_per_rank(self) -> int:
+        n = len(self.dataset)
+        if self.world_size > 1:
+            n = (n + self.world_size - 1) // self.world_size
+        return n
+
     def __len__(self) -> int:
-        n = len(self.dataset) // self.batch_size
-        if not self.drop_last and len(self.dataset) % self.batch_size:
+        n_items = self._items_per_rank()
+        n = n_items // self.batch_size
+        if not self.drop_last and n_items % self.batch_size:
             n += 1
         return n
 
@@ -52,7 +69,8 @@ class StatefulTaskDataLoader:
                 break
             batch = [self.dataset[i] for i in order[lo:hi]]
             self.batch_idx += 1
-            yield batch
+            if batch:
+                yield batch
         self.epoch += 1
         self.batch_idx = 0
 

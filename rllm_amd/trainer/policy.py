@@ -292,7 +292,7 @@ class PolicyTrainer:
             if overlap and mi + 1 < len(packed) and (mi + 1) not in prefetched:
                 with torch.cuda.stream(ref_stream):
                     prefetched[mi + 1] = ref_lp_of(mi + 1)
-            tot_loss += float(loss.detach()) * (1.0 if cfg.loss_agg_mode != "token-mean" else 1.0)
+            tot_loss += float(loss.detach())
             tot_pg += float(loss_tok.detach().sum())
             tot_clip += float(clipped.sum())
             tot_kl_count += int(loss_tok.numel())

@@ -273,13 +273,19 @@ class UnifiedTrainer:
             val_tasks = val_tasks[: cfg.val_batch_size]
         tasks, uids = interleave_tasks(val_tasks, cfg.rollout_n_val)
         episodes = await self.backend.generate_episodes(tasks, uids=uids, is_validation=True)
-        n = len(episodes) or 1
-        acc = sum(1 for e in episodes if e.is_correct) / n
+        # DP: each rank validated its task shard — aggregate globally so
+        # rank 0's logged numbers cover the whole val set
+        from rllm_amd.parallel import dist as pdist
+
+        n_correct = pdist.all_reduce_scalar(float(sum(1 for e in episodes if e.is_correct)))
+        n_eps = pdist.all_reduce_scalar(float(len(episodes)))
         rewards = [t.reward or 0.0 for e in episodes for t in e.trajectories]
+        r_sum = pdist.all_reduce_scalar(float(sum(rewards)))
+        r_n = pdist.all_reduce_scalar(float(len(rewards)))
         metrics = {
-            "val/accuracy": acc,
-            "val/reward_mean": sum(rewards) / max(1, len(rewards)),
-            "val/num_episodes": len(episodes),
+            "val/accuracy": n_correct / max(1.0, n_eps),
+            "val/reward_mean": r_sum / max(1.0, r_n),
+            "val/num_episodes": n_eps,
         }
         if self.episode_logger:
             self.episode_logger.log_episodes(episodes, mode="val",

@@ -391,3 +391,24 @@ def _worker_global_batch_gather(rank, world, q):
 @pytest.mark.timeout(180)
 def test_global_batch_gather_world2():
     _run_spawn(_worker_global_batch_gather, "29623")
+
+
+def test_dataloader_rank_shards_are_lockstep():
+    """Every rank must see the SAME batch count per epoch (uneven shards
+    would diverge global_step and deadlock the DP collectives)."""
+    from rllm_amd.data.dataloader import StatefulTaskDataLoader
+
+    for n, world, bs in [(5, 2, 2), (7, 4, 2), (8, 8, 1), (13, 8, 2), (3, 8, 2)]:
+        counts = []
+        seen = set()
+        for rank in range(world):
+            dl = StatefulTaskDataLoader(list(range(n)), batch_size=bs, seed=1,
+                                        rank=rank, world_size=world)
+            batches = list(dl)
+            assert len(batches) == len(dl), (n, world, bs, rank)
+            assert all(batches), f"empty batch at n={n} world={world} rank={rank}"
+            counts.append(len(batches))
+            for b in batches:
+                seen.update(b)
+        assert len(set(counts)) == 1, (n, world, bs, counts)
+        assert seen == set(range(n)), "some items never visited"
