@@ -266,17 +266,22 @@ class UnifiedTrainer:
 
     # ------------------------------------------------------------------
     async def _validate(self) -> dict:
+        from rllm_amd.parallel import dist as pdist
+
         cfg = self.config
         self.backend.on_validation_start()
         val_tasks = list(self.val_dataset)
         if cfg.val_batch_size:
             val_tasks = val_tasks[: cfg.val_batch_size]
+        # DP: shard the val set (metrics are all-reduced below); every rank
+        # participates even with an empty shard — the collectives must match
+        world = pdist.get_world_size()
+        if world > 1:
+            val_tasks = val_tasks[pdist.get_rank():: world]
         tasks, uids = interleave_tasks(val_tasks, cfg.rollout_n_val)
         episodes = await self.backend.generate_episodes(tasks, uids=uids, is_validation=True)
         # DP: each rank validated its task shard — aggregate globally so
         # rank 0's logged numbers cover the whole val set
-        from rllm_amd.parallel import dist as pdist
-
         n_correct = pdist.all_reduce_scalar(float(sum(1 for e in episodes if e.is_correct)))
         n_eps = pdist.all_reduce_scalar(float(len(episodes)))
         rewards = [t.reward or 0.0 for e in episodes for t in e.trajectories]
