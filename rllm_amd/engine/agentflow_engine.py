@@ -141,6 +141,13 @@ class AgentFlowEngine:
         self._accepts_env = flow_accepts_env(agent_flow)
         self.is_validation = False
 
+    def close(self) -> None:
+        try:
+            self.http.close()
+        except Exception:  # noqa: BLE001
+            pass
+        self.executor.shutdown(wait=False, cancel_futures=True)
+
     # ------------------------------------------------------------------
     async def execute_tasks(self, tasks: list[Task], task_ids: list[str] | None = None,
                             is_validation: bool = False) -> list[Episode]:

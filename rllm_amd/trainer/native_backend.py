@@ -252,6 +252,8 @@ class NativeBackend(BackendProtocol):
         return int(sd.get("weight_version", 0))
 
     def shutdown(self) -> None:
+        if self.flow_engine is not None:
+            self.flow_engine.close()
         if self.gateway is not None:
             self.gateway.stop()
         if self.driver is not None:
