@@ -58,6 +58,38 @@ class ChatTemplateParser:
             prev_len = len(msgs)
         return ids, mask
 
+    def verify_equivalence(self, sample_messages: list[dict] | None = None) -> bool:
+        """Check this parser renders identically to the tokenizer's OWN
+        chat template when one exists (reference chat_template_parser.py:50
+        — a silently-divergent parser corrupts every training mask).
+        Returns True when equivalent or unverifiable; logs a loud warning
+        on divergence."""
+        apply = getattr(self.tokenizer, "apply_chat_template", None)
+        if apply is None or not getattr(self.tokenizer, "chat_template", None):
+            return True
+        msgs = sample_messages or [
+            {"role": "system", "content": "You are helpful."},
+            {"role": "user", "content": "What is 2+2?"},
+            {"role": "assistant", "content": "4"},
+            {"role": "user", "content": "And 3+3?"},
+        ]
+        try:
+            theirs = apply(msgs, tokenize=False, add_generation_prompt=True)
+        except Exception:  # noqa: BLE001 — template needs extra context; skip
+            return True
+        ours = self.format(msgs, add_generation_prompt=True)
+        if ours != theirs:
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "chat parser diverges from the tokenizer's own template!\n"
+                "parser renders: %r\ntemplate renders: %r\n"
+                "Training masks would be wrong — pick the right parser via "
+                "ChatTemplateParser.get_parser(tokenizer, model_family).",
+                ours[:200], theirs[:200])
+            return False
+        return True
+
     @staticmethod
     def get_parser(tokenizer, model_family: str = "qwen") -> "ChatTemplateParser":
         fam = (model_family or "qwen").lower()

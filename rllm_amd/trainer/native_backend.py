@@ -89,6 +89,11 @@ class NativeBackend(BackendProtocol):
 
             tokenizer = load_tokenizer(self.hf_model_dir)
         self.tokenizer = tokenizer or ByteTokenizer()
+        if parser is None and self.hf_model_dir is not None:
+            # real tokenizer: dispatch by model family and verify the
+            # rendering against the tokenizer's own chat template
+            parser = ChatTemplateParser.get_parser(self.tokenizer, self.cfg.name)
+            parser.verify_equivalence()
         self.parser = parser or QwenChatTemplateParser(self.tokenizer)
         self.policy_config = policy_config or PolicyTrainerConfig()
         self.use_ref = use_ref and self.policy_config.kl_beta > 0

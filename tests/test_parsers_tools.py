@@ -207,3 +207,35 @@ def test_r1_tool_parser_special_tokens():
     from rllm_amd.parser.tool_parser import QwenToolParser
 
     assert "<tool_call>" in QwenToolParser().get_tool_prompt("{}")
+
+
+def test_verify_equivalence_detects_divergence(caplog):
+    """verify_equivalence flags a parser whose rendering differs from the
+    tokenizer's own chat template (reference :50)."""
+    try:
+        from tokenizers import Tokenizer, models
+        from transformers import PreTrainedTokenizerFast
+    except ImportError:
+        import pytest
+
+        pytest.skip("transformers/tokenizers unavailable")
+
+    from rllm_amd.parser.chat_template_parser import (
+        LlamaChatTemplateParser,
+        QwenChatTemplateParser,
+    )
+
+    tpl = ("{% for message in messages %}"
+           "{{'<|im_start|>' + message['role'] + '\n' + message['content'] + '<|im_end|>' + '\n'}}"
+           "{% endfor %}"
+           "{% if add_generation_prompt %}{{ '<|im_start|>assistant\n' }}{% endif %}")
+    hf_tok = PreTrainedTokenizerFast(
+        tokenizer_object=Tokenizer(models.WordLevel({"[UNK]": 0}, unk_token="[UNK]")),
+        chat_template=tpl)
+
+    assert QwenChatTemplateParser(hf_tok).verify_equivalence() is True
+    assert LlamaChatTemplateParser(hf_tok).verify_equivalence() is False
+    # tokenizers without a template are unverifiable -> True
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    assert QwenChatTemplateParser(ByteTokenizer()).verify_equivalence() is True
