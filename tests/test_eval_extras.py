@@ -52,3 +52,90 @@ def test_curation_to_sft_rows():
     # dedupe identical transcripts
     rows2 = episodes_to_sft_rows([make_ep("x", 0, 1.0, True), make_ep("x", 1, 1.0, True)])
     assert len(rows2) == 1
+
+
+def test_provider_proxy_routing_and_translation():
+    """Provider proxy (reference _litellm_server role): model-prefix
+    routing, key injection, Anthropic request/response translation."""
+    from rllm_amd.eval.provider_proxy import ProviderProxy
+
+    seen = []
+
+    def fake(url, payload, headers):
+        seen.append((url, payload, headers))
+        if "/v1/messages" in url:
+            return 200, {"id": "m1", "content": [{"type": "text", "text": "claude says hi"}],
+                         "stop_reason": "end_turn",
+                         "usage": {"input_tokens": 7, "output_tokens": 3}}
+        return 200, {"id": "c1", "object": "chat.completion", "model": payload["model"],
+                     "choices": [{"index": 0, "message": {"role": "assistant", "content": "ok"},
+                                  "finish_reason": "stop"}]}
+
+    proxy = ProviderProxy({
+        "openai": {"base_url": "https://api.openai.example/v1", "api_key": "sk-o"},
+        "anthropic": {"base_url": "https://api.anthropic.example",
+                      "api_key": "sk-a", "style": "anthropic"},
+    }, transport=fake)
+
+    # openai passthrough: prefix stripped, bearer injected
+    status, resp = proxy.handle_chat({"model": "openai/gpt-x",
+                                      "messages": [{"role": "user", "content": "hi"}]})
+    assert status == 200 and resp["choices"][0]["message"]["content"] == "ok"
+    url, payload, headers = seen[-1]
+    assert url.endswith("/chat/completions") and payload["model"] == "gpt-x"
+    assert headers["Authorization"] == "Bearer sk-o"
+
+    # anthropic: translated both ways, system lifted out of messages
+    status, resp = proxy.handle_chat({
+        "model": "anthropic/claude-x", "max_tokens": 32, "temperature": 0.5,
+        "messages": [{"role": "system", "content": "be brief"},
+                     {"role": "user", "content": "hi"}]})
+    url, payload, headers = seen[-1]
+    assert url.endswith("/v1/messages") and headers["x-api-key"] == "sk-a"
+    assert payload["system"] == "be brief" and payload["temperature"] == 0.5
+    assert all(m["role"] != "system" for m in payload["messages"])
+    assert status == 200
+    assert resp["object"] == "chat.completion"
+    assert resp["choices"][0]["message"]["content"] == "claude says hi"
+    assert resp["choices"][0]["finish_reason"] == "stop"
+    assert resp["usage"]["prompt_tokens"] == 7
+
+    # unprefixed model falls to the default provider
+    proxy.handle_chat({"model": "bare-model", "messages": []})
+    assert seen[-1][1]["model"] == "bare-model"
+
+
+def test_provider_proxy_http_server():
+    """The proxy serves a real OpenAI-compatible endpoint."""
+    import httpx
+
+    from rllm_amd.eval.provider_proxy import ProviderProxy
+
+    def fake(url, payload, headers):
+        return 200, {"id": "c", "object": "chat.completion", "model": payload["model"],
+                     "choices": [{"index": 0, "message": {"role": "assistant", "content": "pong"},
+                                  "finish_reason": "stop"}]}
+
+    proxy = ProviderProxy({"local": {"base_url": "http://up/v1"}}, transport=fake)
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    base = proxy.start(port=port)
+    try:
+        import time
+
+        for _ in range(50):
+            try:
+                r = httpx.get(base.replace("/v1", "/health"), timeout=1.0)
+                break
+            except Exception:  # noqa: BLE001
+                time.sleep(0.1)
+        r = httpx.post(base + "/chat/completions",
+                       json={"model": "m", "messages": [{"role": "user", "content": "ping"}]},
+                       timeout=10.0)
+        assert r.status_code == 200
+        assert r.json()["choices"][0]["message"]["content"] == "pong"
+    finally:
+        proxy.stop()
