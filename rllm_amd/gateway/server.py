@@ -15,6 +15,7 @@ OpenAI-compatible reverse proxy with per-session trace capture:
   POST /admin/workers               register worker {url, model}
   DELETE /admin/workers             {url}
   POST /admin/flush                 clear traces
+  POST /admin/reload                reset router health/stickiness
   GET/POST /admin/weight_version    version stamp for async staleness
   ANY  /sessions/{sid}/v1/{path}    traced proxy (the agent-facing base_url)
   ANY  /v1/{path}                   untraced proxy
@@ -154,6 +155,17 @@ def create_app(config: GatewayConfig | None = None,
     async def flush():
         await st.store.flush()
         return {"status": "flushed"}
+
+    @app.post("/admin/reload")
+    async def reload_workers():
+        # reference server.py /admin/reload: re-probe worker health and drop
+        # session stickiness so rebalancing starts fresh after a fleet change
+        await st.registry.health_check(st.http)
+        st.policy._session_worker.clear()
+        for w in st.registry.list():
+            w.active_sessions = 0
+        return {"status": "reloaded",
+                "workers": [w.model_dump() for w in st.registry.list()]}
 
     @app.get("/admin/weight_version")
     async def get_weight_version():

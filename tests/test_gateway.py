@@ -318,3 +318,19 @@ def test_token_accumulator_invariant_random_turns():
         assert acc.build_prompt_ids(sid, msgs[:1]) is None
         acc.reset(sid)
         assert acc.build_prompt_ids(sid, msgs) is None
+
+
+def test_admin_reload_resets_stickiness(stack):
+    gw, worker = stack
+    import httpx
+
+    base = gw.base_url
+    httpx.post(base + "/sessions", json={"session_id": "sticky-1"})
+    httpx.post(base + "/sessions/sticky-1/v1/chat/completions",
+               json={"model": "m", "messages": [{"role": "user", "content": "x"}]},
+               timeout=30.0)
+    r = httpx.post(base + "/admin/reload", timeout=10.0)
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "reloaded" and body["workers"]
+    assert all(w["active_sessions"] == 0 for w in body["workers"])
