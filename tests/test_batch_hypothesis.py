@@ -83,3 +83,56 @@ def test_pack_rows_positions_restart_per_sequence(rows):
         assert seg == list(range(len(row.tokens)))
     assert batch.input_ids.dtype == torch.long
     assert batch.positions.dtype == torch.int32
+
+
+@given(st.lists(
+    st.tuples(st.integers(2, 40),                      # seq length
+              st.integers(0, 39)),                     # prompt length (capped below)
+    min_size=1, max_size=8))
+@settings(max_examples=60, deadline=None)
+def test_pack_rows_matches_slow_reference(shapes):
+    """The vectorized pack must agree with a per-token reference packer."""
+    import numpy as np
+
+    from rllm_amd.trainer.batch import PackedRow, pack_rows
+
+    rng = np.random.default_rng(0)
+    rows = []
+    for n, p in shapes:
+        p = min(p, n - 1)
+        toks = rng.integers(0, 1000, n).tolist()
+        mask = [0] * p + [1] * (n - p)
+        rows.append(PackedRow(tokens=toks, response_mask=mask,
+                              advantages=rng.standard_normal(n).tolist(),
+                              rollout_logprobs=rng.standard_normal(n).tolist()))
+
+    b = pack_rows(rows)
+
+    # slow reference
+    ids, lm, tgt, adv, rl, cu, n_resp = [], [], [], [], [], [0], 0
+    for row in rows:
+        n = len(row.tokens)
+        ids.extend(row.tokens)
+        cu.append(cu[-1] + n)
+        for r in range(n):
+            if r + 1 < n and row.response_mask[r + 1]:
+                lm.append(True)
+                tgt.append(row.tokens[r + 1])
+                adv.append(row.advantages[r + 1])
+                rl.append(row.rollout_logprobs[r + 1])
+                n_resp += 1
+            else:
+                lm.append(False)
+                tgt.append(0)
+                adv.append(0.0)
+                rl.append(0.0)
+
+    assert b.input_ids.tolist() == ids
+    assert b.cu_seqlens == cu
+    assert b.loss_mask.tolist() == lm
+    assert b.targets.tolist() == tgt
+    assert b.n_response_tokens == n_resp
+    import torch
+
+    assert torch.allclose(b.advantages, torch.tensor(adv, dtype=torch.float32))
+    assert torch.allclose(b.rollout_logprobs, torch.tensor(rl, dtype=torch.float32))
