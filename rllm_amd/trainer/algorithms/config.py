@@ -119,6 +119,9 @@ class RejectionSamplingConfig:
     min_trajs_per_group: int = 2
     min_partial_solve_tasks: int = 1
     filter_uniform_groups: bool = False
+    # group mode: accumulate FILTERED groups across batches and release
+    # once this many informative groups exist (0 = release every batch)
+    min_groups_per_batch: int = 0
 
     @classmethod
     def from_config(cls, config) -> "RejectionSamplingConfig":
@@ -130,6 +133,7 @@ class RejectionSamplingConfig:
             min_trajs_per_group=int(_get(config, "min_trajs_per_group", 2)),
             min_partial_solve_tasks=int(_get(config, "min_partial_solve_tasks", 1)),
             filter_uniform_groups=bool(_get(config, "filter_uniform_groups", False)),
+            min_groups_per_batch=int(_get(config, "min_groups_per_batch", 0)),
         )
 
 

@@ -118,9 +118,6 @@ def apply_rejection_sampling_and_filtering(
 ) -> tuple[list[TrajectoryGroup], list[Episode], dict]:
     """Apply rejection sampling; in episode mode accumulate until
     min_partial_solve_tasks is reached (returns empty lists otherwise)."""
-    if config.mode == "group":
-        raise NotImplementedError("Group-level rejection sampling is not implemented yet")
-
     metrics = state.metrics
     filtered_groups, dropped_groups = filter_groups(groups, config, metrics)
     filtered_episodes = filter_episodes(episodes, dropped_groups)
@@ -132,6 +129,16 @@ def apply_rejection_sampling_and_filtering(
         state.accumulated_groups.extend(filtered_groups)
         state.accumulated_episodes.extend(filtered_episodes)
         if metrics.solve_partial >= config.min_partial_solve_tasks:
+            return state.accumulated_groups.copy(), state.accumulated_episodes.copy(), metrics.to_dict()
+        return [], [], metrics.to_dict()
+    if config.mode == "group":
+        # group-level accumulation (a stub in the reference,
+        # rejection_sampling.py:182 — implemented here): keep only
+        # informative groups and release once min_groups_per_batch exist,
+        # so every optimizer step sees a full batch of learning signal
+        state.accumulated_groups.extend(filtered_groups)
+        state.accumulated_episodes.extend(filtered_episodes)
+        if len(state.accumulated_groups) >= max(1, config.min_groups_per_batch):
             return state.accumulated_groups.copy(), state.accumulated_episodes.copy(), metrics.to_dict()
         return [], [], metrics.to_dict()
     raise ValueError(f"Unknown rejection sampling mode: {config.mode}")

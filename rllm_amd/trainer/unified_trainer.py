@@ -214,7 +214,7 @@ class UnifiedTrainer:
         if n_groups_global <= 0:
             metrics["batch/skipped"] = 1.0
             return metrics
-        if groups and self.rs_config.mode == "episode":
+        if groups and self.rs_config.mode in ("episode", "group"):
             self.rs_state.reset()
 
         # C4 — optional trajectory all-gather: every rank sees the GLOBAL

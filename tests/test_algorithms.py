@@ -208,3 +208,41 @@ def test_advantage_properties_random():
         g, _ = calculate_grpo_advantages_per_group(uniform)
         l, _ = calculate_rloo_advantages_per_group(uniform)
         assert np.all(np.abs(g) < 1e-4) and np.all(np.abs(l) < 1e-6)
+
+
+def test_group_mode_rejection_sampling_accumulates():
+    """Group mode (a NotImplementedError stub in the reference): keep only
+    informative groups, accumulate across batches, release at
+    min_groups_per_batch."""
+    from rllm_amd.trainer.algorithms.config import RejectionSamplingConfig
+    from rllm_amd.trainer.algorithms.rejection_sampling import (
+        RejectionSamplingState,
+        apply_rejection_sampling_and_filtering,
+    )
+    from rllm_amd.types import Episode, Step, Trajectory, TrajectoryGroup
+
+    def make(task_id, rewards):
+        eps, trajs = [], []
+        for i, r in enumerate(rewards):
+            st = Step(prompt_ids=[1, 2], response_ids=[3, 4], logprobs=[-0.1, -0.1],
+                      reward=r, done=True)
+            tr = Trajectory(name="s", steps=[st], reward=r)
+            trajs.append(tr)
+            eps.append(Episode(id=f"{task_id}:{i}", trajectories=[tr], is_correct=r > 0))
+        return eps, TrajectoryGroup(group_id=f"{task_id}:s", trajectories=trajs)
+
+    cfg = RejectionSamplingConfig(mode="group", filter_uniform_groups=True,
+                                  min_groups_per_batch=2)
+    state = RejectionSamplingState()
+
+    # batch 1: one informative group + one uniform (dropped) -> accumulate
+    e1, g1 = make("a", [0.0, 1.0])
+    e2, g2 = make("b", [1.0, 1.0])
+    groups, eps, m = apply_rejection_sampling_and_filtering(e1 + e2, [g1, g2], cfg, state)
+    assert groups == [] and len(state.accumulated_groups) == 1
+
+    # batch 2: another informative group -> release both
+    e3, g3 = make("c", [0.0, 1.0])
+    groups, eps, m = apply_rejection_sampling_and_filtering(e3, [g3], cfg, state)
+    assert [g.group_id for g in groups] == ["a:s", "c:s"]
+    assert m["batch/groups_after_filter"] == 2
