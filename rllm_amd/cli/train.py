@@ -40,9 +40,16 @@ def resolve_ref(ref: str):
 @click.option("--logger", "loggers", multiple=True, default=["console"])
 @click.option("--lora-rank", default=0, type=int, help="train LoRA adapters of this rank (0 = full finetune)")
 @click.option("--cumulative", is_flag=True, help="gateway cumulative token mode (multi-turn agents)")
+@click.option("--global-batch", is_flag=True,
+              help="C4: all-gather episodes across DP ranks + token-balanced shard")
+@click.option("--context-curriculum", default=None,
+              help="DeepScaleR schedule 'step:max_tokens,step:max_tokens' e.g. '0:8192,200:16384,400:24576'")
+@click.option("--teacher-url", default=None, help="on-policy distillation teacher /v1 endpoint")
+@click.option("--teacher-model", default=None)
 def train(dataset, agent_ref, evaluator_ref, backend, split, epochs, batch_size, rollout_n,
           max_steps, lr, kl_beta, estimator, model_config, checkpoint_dir, save_freq,
-          registry_dir, loggers, lora_rank, cumulative):
+          registry_dir, loggers, lora_rank, cumulative, global_batch, context_curriculum,
+          teacher_url, teacher_model):
     """Train an agent with GRPO/RLOO on a registered dataset."""
     from rllm_amd.data.dataset import DatasetRegistry
     from rllm_amd.trainer.agent_trainer import AgentTrainer
@@ -65,12 +72,23 @@ def train(dataset, agent_ref, evaluator_ref, backend, split, epochs, batch_size,
         from rllm_amd.gateway.models import GatewayConfig
 
         backend_kwargs["gateway_config"] = GatewayConfig(cumulative_token_mode=True)
+    if teacher_url:
+        from rllm_amd.trainer.distill import TeacherClient
+
+        backend_kwargs["distill"] = {"teacher": TeacherClient(teacher_url, teacher_model or "")}
+
+    curriculum = None
+    if context_curriculum:
+        curriculum = [[int(a), int(b)] for a, b in
+                      (pair.split(":") for pair in context_curriculum.split(","))]
 
     cfg = TrainerConfig(
         total_epochs=epochs, train_batch_size=batch_size, rollout_n=rollout_n,
         max_steps=max_steps, checkpoint_dir=checkpoint_dir, save_freq=save_freq,
-        logger_backends=list(loggers))
-    algo = AlgorithmConfig(estimator=estimator, kl_beta=kl_beta)
+        logger_backends=list(loggers), global_batch_mode=global_batch,
+        context_curriculum=curriculum)
+    algo = AlgorithmConfig(estimator=estimator, kl_beta=kl_beta,
+                           use_precomputed_advantage=bool(teacher_url))
 
     AgentTrainer(
         agent_flow=flow, evaluator=evaluator, train_dataset=tasks,
