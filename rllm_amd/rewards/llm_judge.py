@@ -12,6 +12,7 @@ from __future__ import annotations
 
 import json
 import logging
+import os
 import re
 from typing import Callable
 
@@ -158,3 +159,56 @@ def make_equality_judge(base_url: str, model: str, timeout: float = 60.0):
             return None
 
     return judge
+
+
+def trajectory_transcript(episode, max_chars: int = 12000) -> str:
+    """Flatten an episode's full conversation for a trajectory-level judge."""
+    parts = []
+    for traj in episode.trajectories:
+        for st in traj.steps:
+            for m in st.chat_completions or []:
+                parts.append(f"{m.get('role', '?')}: {m.get('content', '')}")
+            if st.model_response and (not st.chat_completions
+                                      or st.chat_completions[-1].get("role") != "assistant"):
+                parts.append(f"assistant: {st.model_response}")
+    text = "\n".join(parts)
+    if len(text) > max_chars:
+        text = text[: max_chars // 2] + "\n...[truncated]...\n" + text[-max_chars // 2:]
+    return text
+
+
+def claw_eval_reward_fn(task, episode):
+    """Trajectory-completion judge (reference claw_eval.py: the public
+    Claw-Eval rows ship no machine rubrics, so grading is an LLM judgement
+    of whether the agent ACCOMPLISHED the task, over the full transcript).
+    Never raises — judge-unreachable returns reward 0 with a reason signal
+    so plumbing failures are distinguishable from agent failures."""
+    from rllm_amd.eval.types import EvalOutput
+
+    md = getattr(task, "metadata", None) or {}
+    query = md.get("query") or str(getattr(task, "instruction", "") or "")
+    base_url = md.get("judge_base_url") or os.environ.get("RLLM_JUDGE_BASE_URL")
+    model = md.get("judge_model") or os.environ.get("RLLM_JUDGE_MODEL", "")
+    chat = md.get("judge_chat_fn")  # injectable for tests
+    if chat is None:
+        if not base_url:
+            return EvalOutput(reward=0.0, is_correct=False,
+                              signals={"judge_unavailable": 1.0})
+        chat = http_chat_fn(base_url, model)
+
+    transcript = trajectory_transcript(episode)
+    prompt = (
+        "You are grading an AI assistant's attempt at a task.\n\n"
+        f"TASK:\n{query}\n\nFULL TRANSCRIPT:\n{transcript}\n\n"
+        "Did the assistant ACCOMPLISH the task? Judge the outcome, not the "
+        'style. Respond with JSON only: {"accomplished": true/false, '
+        '"reason": "<one sentence>"}')
+    try:
+        out = chat([{"role": "user", "content": prompt}])
+        verdict = json.loads(out[out.index("{"): out.rindex("}") + 1])
+        ok = bool(verdict.get("accomplished"))
+        return EvalOutput(reward=1.0 if ok else 0.0, is_correct=ok,
+                          signals={"judge_accomplished": 1.0 if ok else 0.0})
+    except Exception:  # noqa: BLE001 — judge failure is not an agent failure
+        return EvalOutput(reward=0.0, is_correct=False,
+                          signals={"judge_error": 1.0})

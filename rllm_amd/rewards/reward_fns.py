@@ -166,6 +166,7 @@ REWARD_FN_REGISTRY = {
     "point_in_mask": "rllm_amd.rewards.extra_rewards:point_in_mask_reward_fn",
     "search": "rllm_amd.rewards.extra_rewards:search_reward_fn",
     "widesearch": "rllm_amd.rewards.extra_rewards:widesearch_reward_fn",
+    "claw_eval": "rllm_amd.rewards.llm_judge:claw_eval_reward_fn",
 }
 
 

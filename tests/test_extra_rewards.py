@@ -204,3 +204,38 @@ def test_code_reward_unshare_isolation():
             "    finally:\n        s.close()\n```",
             [{"assert": "assert f(2) == 4"}])
         assert out.is_correct  # connect refused/unreachable inside the netns
+
+
+def test_claw_eval_trajectory_judge():
+    """claw_eval grader: full-transcript completion judgement with an
+    injectable judge; judge failures are signals, never exceptions."""
+    from rllm_amd.rewards.llm_judge import claw_eval_reward_fn, trajectory_transcript
+
+    ep = _episode("I created the file as requested.")
+    t_ok = Task(id="1", instruction="create a file", metadata={
+        "judge_chat_fn": lambda msgs: '{"accomplished": true, "reason": "done"}'})
+    out = claw_eval_reward_fn(t_ok, ep)
+    assert out.reward == 1.0 and out.is_correct
+
+    t_no = Task(id="1", instruction="create a file", metadata={
+        "judge_chat_fn": lambda msgs: 'verdict: {"accomplished": false, "reason": "nope"}'})
+    out = claw_eval_reward_fn(t_no, ep)
+    assert out.reward == 0.0 and not out.is_correct
+
+    # transcript includes the conversation
+    assert "assistant" in trajectory_transcript(ep)
+
+    # no judge configured -> finite reward with a diagnostic signal
+    import os
+
+    os.environ.pop("RLLM_JUDGE_BASE_URL", None)
+    out = claw_eval_reward_fn(Task(id="1", instruction="x", metadata={}), ep)
+    assert out.reward == 0.0 and out.signals.get("judge_unavailable") == 1.0
+
+    # judge crash -> judge_error signal, no exception
+    def boom(msgs):
+        raise RuntimeError("judge down")
+
+    out = claw_eval_reward_fn(Task(id="1", instruction="x",
+                                   metadata={"judge_chat_fn": boom}), ep)
+    assert out.signals.get("judge_error") == 1.0
