@@ -205,3 +205,67 @@ def test_tool_calling_flow_e2e():
     assert "The answer is 42" in steps[1].model_response
     for st in steps:
         assert st.response_ids and len(st.logprobs) == len(st.response_ids)
+
+
+def test_tool_loop_with_cumulative_token_mode():
+    """Multi-turn tool agent THROUGH cumulative token mode: turn N+1's
+    prompt ids must prefix-extend turn N's (the training-mask invariant),
+    with the tool result appearing as an observation segment."""
+    import asyncio
+    import json as _json
+
+    from rllm_amd.engine.agentflow_engine import AgentFlowEngine
+    from rllm_amd.gateway.manager import GatewayManager
+    from rllm_amd.gateway.models import GatewayConfig
+    from rllm_amd.gateway.native_adapter import make_torch_lm_local_handler
+    from rllm_amd.models.torch_lm import TinyTorchLM
+    from rllm_amd.parser.chat_template_parser import QwenChatTemplateParser
+    from rllm_amd.tools.python_tool import CalculatorTool
+    from rllm_amd.trainer.batch import rows_from_episodes
+    from rllm_amd.types import Task
+    from rllm_amd.utils.tokenizer import ByteTokenizer
+
+    parser = QwenChatTemplateParser(ByteTokenizer())
+    model = TinyTorchLM(seed=0)
+    handler = make_torch_lm_local_handler(model, parser, eos_token_id=None, seed=0)
+
+    # wrap: first turn emits a tool call TEXT, later turns delegate to the model
+    turn_count = {"n": 0}
+    call_txt = ('<tool_call>\n{"name": "calculator", "arguments": {"expression": "6*7"}}\n'
+                "</tool_call>")
+
+    async def scripted(request):
+        resp = await handler(request)
+        turn_count["n"] += 1
+        if turn_count["n"] == 1:
+            # keep token ids AND text consistent: re-encode the tool-call text
+            toks = parser.tokenizer.encode(call_txt)
+            choice = resp["choices"][0]
+            choice["message"]["content"] = call_txt
+            choice["token_ids"] = toks
+            choice["logprobs"] = {"token_logprobs": [-0.1] * len(toks)}
+        return resp
+
+    gw = GatewayManager(GatewayConfig(cumulative_token_mode=True), local_handler=scripted,
+                        parser=parser)
+    gw.start()
+    try:
+        from rllm_amd.workflows.tool_loop import ToolCallingFlow
+
+        flow = ToolCallingFlow([CalculatorTool()], max_turns=3)
+        engine = AgentFlowEngine(flow, gw, evaluator=None, n_parallel_tasks=1)
+        eps = asyncio.run(engine.execute_tasks([Task(id="t", instruction="6*7?")], ["t:0"]))
+    finally:
+        gw.stop()
+
+    ep = eps[0]
+    steps = ep.trajectories[0].steps
+    assert len(steps) >= 2
+    # cumulative invariant: turn 2's prompt ids prefix-extend turn 1's
+    full1 = steps[0].prompt_ids + steps[0].response_ids
+    assert steps[1].prompt_ids[: len(full1)] == full1
+    # and the merged training row masks the tool observation as context
+    rows = rows_from_episodes(eps)
+    assert len(rows) == 1  # prefix merge collapsed the turns into one row
+    row = rows[0]
+    assert sum(row.response_mask) == sum(len(s.response_ids) for s in steps)
