@@ -10,7 +10,8 @@ from rllm_amd.cli.main import cli
 def test_cli_help_lists_commands():
     r = CliRunner().invoke(cli, ["--help"])
     assert r.exit_code == 0
-    for cmd in ("train", "eval", "dataset", "view", "bench", "build-kernels"):
+    for cmd in ("train", "eval", "dataset", "view", "bench", "build-kernels",
+                "sft", "model", "init", "snapshot", "ui", "serve"):
         assert cmd in r.output
 
 
