@@ -132,3 +132,37 @@ def test_step_profile_region_cpu_noop():
         assert active2 is False
     with roctx_range("x"):
         pass
+
+
+def test_tracking_episode_streaming_fanout():
+    """Tracking.log_episodes reaches every backend exposing log_episode."""
+    from rllm_amd.types import Episode, Step, Trajectory
+    from rllm_amd.utils.tracking import Tracking
+
+    class Sink:
+        def __init__(self):
+            self.metrics = []
+            self.episodes = []
+
+        def log(self, m, step):
+            self.metrics.append((step, m))
+
+        def log_episode(self, d, step):
+            self.episodes.append((step, d))
+
+        def finish(self):
+            pass
+
+    t = Tracking(backends=[])
+    sink = Sink()
+    t.loggers.append(sink)
+    ep = Episode(id="e:0", trajectories=[Trajectory(name="s", steps=[
+        Step(prompt_ids=[1], response_ids=[2], logprobs=[-0.1])])])
+    t.log({"a": 1.0}, step=3)
+    t.log_episodes([ep], step=3)
+    assert sink.metrics == [(3, {"a": 1.0})]
+    assert sink.episodes[0][0] == 3
+    assert sink.episodes[0][1]["id"] == "e:0"
+    # backends without log_episode are skipped silently
+    t2 = Tracking(backends=["console"])
+    t2.log_episodes([ep], step=0)  # no raise
