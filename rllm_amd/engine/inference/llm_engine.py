@@ -505,8 +505,8 @@ class LLMEngine:
             positions.append(pos)
             slot_mapping.append(self._slot(seq, pos))
             seq_lens.append(pos + 1)
-        if not active:
-            return 0
+        if not active or any(s.state != SeqState.RUNNING for s in active):
+            return 0  # late preemption invalidated an earlier-staged seq
         batch = active
         B = len(batch)
         max_pages = max(len(s.pages) for s in batch)
@@ -670,6 +670,12 @@ class LLMEngine:
             sched_np[:, B:Bp] = 0
             len_np[B:Bp] = 1
             bt_np[B:Bp, 0] = 0
+
+        # a LATER seq's page grab may have preempted an EARLIER, already-
+        # staged seq — its staged slots would point at freed pages. Bail
+        # and retry with the new batch composition.
+        if any(s.state != SeqState.RUNNING for s in batch):
+            return 0
 
         def stage():
             for k in ("tok", "pos", "len"):
